@@ -1,0 +1,132 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Nonblocking-op handles.
+
+Reference analog: bluefog/torch/handle_manager.{h,cc} plus the spin-wait in
+mpi_ops.cc:549-555. Here a handle wraps stream-ordered work instead of a
+background-thread status slot:
+
+- GPU ops record a hipEvent on the side stream after the post-communication
+  kernel; ``synchronize`` makes the *caller's* stream wait on that event
+  (host never blocks), ``poll`` queries the event.
+- CPU (gloo) ops keep their ``dist.Work`` list and run the post-op callback
+  lazily on first synchronize/wait.
+"""
+
+import threading
+from typing import Callable, List, Optional
+
+import torch
+
+from bluefog_amd.utils.timeline import timeline
+
+
+class OpHandle:
+    __slots__ = (
+        "id",
+        "name",
+        "works",
+        "finalize",
+        "event",
+        "future",
+        "result",
+        "done",
+        "_lock",
+    )
+
+    def __init__(self, hid: int, name: str):
+        self.id = hid
+        self.name = name
+        self.works: List = []
+        self.finalize: Optional[Callable[[], torch.Tensor]] = None
+        self.event: Optional[torch.cuda.Event] = None
+        self.future = None  # concurrent.futures.Future (CPU window ops)
+        self.result: Optional[torch.Tensor] = None
+        self.done = False
+        self._lock = threading.Lock()
+
+    # -- completion --------------------------------------------------------
+    def poll(self) -> bool:
+        if self.done:
+            return True
+        if self.future is not None and not self.future.done():
+            return False
+        if self.event is not None:
+            return bool(self.event.query())
+        return all(w.is_completed() for w in self.works)
+
+    def synchronize(self) -> torch.Tensor:
+        with self._lock:
+            if not self.done:
+                if self.future is not None:
+                    self.result = self.future.result()
+                if self.event is not None:
+                    # GPU path: stream-order the caller behind the post-op
+                    # kernel; no host blocking.
+                    torch.cuda.current_stream().wait_event(self.event)
+                elif self.future is None:
+                    for w in self.works:
+                        w.wait()
+                    if self.finalize is not None:
+                        self.result = self.finalize()
+                self.done = True
+                timeline().end_activity(self.name)
+        return self.result
+
+    def wait_host(self) -> torch.Tensor:
+        """Fully block the host until the op (incl. post-kernel) retired."""
+        out = self.synchronize()
+        if self.event is not None:
+            self.event.synchronize()
+        return out
+
+
+class HandleManager:
+    def __init__(self):
+        self._lock = threading.Lock()
+        self._next = 0
+        self._handles = {}
+        self._outstanding_names = set()
+
+    def allocate(self, name: str) -> OpHandle:
+        with self._lock:
+            if name in self._outstanding_names:
+                raise ValueError(
+                    f"Duplicated tensor name {name!r}: the same op name was "
+                    "submitted again before the previous one finished "
+                    "(reference DUPLICATE_NAME_ERROR, common.h:181-185)."
+                )
+            hid = self._next
+            self._next += 1
+            h = OpHandle(hid, name)
+            self._handles[hid] = h
+            self._outstanding_names.add(name)
+            return h
+
+    def get(self, hid: int) -> OpHandle:
+        with self._lock:
+            h = self._handles.get(hid)
+        if h is None:
+            raise ValueError(f"Unknown bluefog handle {hid}")
+        return h
+
+    def release(self, hid: int) -> None:
+        with self._lock:
+            h = self._handles.pop(hid, None)
+            if h is not None:
+                self._outstanding_names.discard(h.name)
+
+    def poll(self, hid: int) -> bool:
+        return self.get(hid).poll()
+
+    def synchronize(self, hid: int) -> torch.Tensor:
+        h = self.get(hid)
+        out = h.synchronize()
+        self.release(hid)
+        return out
+
+
+_handle_manager = HandleManager()
+
+
+def handle_manager() -> HandleManager:
+    return _handle_manager

@@ -1,0 +1,473 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Neighbor (graph) communication ops — the heart of decentralized training.
+
+Reference analog: neighbor_allreduce / neighbor_allgather / hierarchical /
+pair_gossip (bluefog/torch/mpi_ops.py:383-945, nccl_controller.cc:509-1204).
+
+MI355X mapping: every neighbor exchange is one batched group of RCCL
+send/recv over xGMI point-to-point links. Under the flagship dynamic
+one-peer Exponential-2 schedule each iteration's single send+recv pair
+lands on a distinct xGMI link (7 per GPU, ~153 GB/s each) with zero
+contention — the reason this beats ring allreduce on this fabric. The
+post-communication weighted average is one hand-written CDNA4 kernel
+(csrc/kernels.hip) on a side stream instead of the reference's chain of
+torch slice ops (mpi_ops.cc:99-164).
+"""
+
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.distributed as dist
+
+from bluefog_amd.ops import engine, hip_ext
+from bluefog_amd.ops.context import ctx
+
+__all__ = [
+    "neighbor_allreduce",
+    "neighbor_allreduce_nonblocking",
+    "neighbor_allgather",
+    "neighbor_allgather_nonblocking",
+    "hierarchical_neighbor_allreduce",
+    "hierarchical_neighbor_allreduce_nonblocking",
+    "pair_gossip",
+    "pair_gossip_nonblocking",
+]
+
+
+def _post_neighbor_exchange(
+    tensor: torch.Tensor,
+    src_ranks: List[int],
+    dst_ranks: List[int],
+    dst_weights: Dict[int, float],
+    dst_weighting_enabled: bool,
+):
+    """Post one batched RCCL send/recv group: recv a same-shaped slice from
+    every src into a contiguous gather buffer, send (optionally pre-scaled)
+    copies to every dst. Returns (works, gathered, keep_alive)."""
+    d0 = tensor.shape[0] if tensor.dim() else 1
+    rest = list(tensor.shape[1:]) if tensor.dim() else []
+    n_src = len(src_ranks)
+    gathered = (
+        tensor.new_empty([n_src * d0] + rest) if n_src else None
+    )
+    keep_alive = [tensor]
+    ops = []
+    for i, src in enumerate(src_ranks):
+        ops.append(dist.P2POp(dist.irecv, gathered.narrow(0, i * d0, d0), src))
+    send_cache: Dict[float, torch.Tensor] = {}
+    for dst in dst_ranks:
+        w = dst_weights.get(dst, 1.0)
+        if dst_weighting_enabled and w != 1.0:
+            st = send_cache.get(w)
+            if st is None:
+                st = tensor.mul(w)
+                send_cache[w] = st
+                keep_alive.append(st)
+        else:
+            st = tensor
+        ops.append(dist.P2POp(dist.isend, st, dst))
+    works = engine.batch_p2p(ops)
+    if gathered is not None:
+        keep_alive.append(gathered)
+    return works, gathered, keep_alive
+
+
+def _neighbor_allreduce_nonblocking_impl(
+    tensor: torch.Tensor,
+    self_weight: Optional[float],
+    src_weights,
+    dst_weights,
+    enable_topo_check: bool,
+    name: Optional[str],
+) -> int:
+    c = ctx()
+    tensor = tensor.detach()
+    if not tensor.is_contiguous():
+        tensor = tensor.contiguous()
+    (
+        self_weight,
+        src_weights,
+        dst_weights,
+        dynamic_enabled,
+        dst_weighting_enabled,
+    ) = engine.resolve_recv_weights(self_weight, src_weights, dst_weights)
+
+    src_ranks = list(src_weights.keys())
+    dst_ranks = list(dst_weights.keys())
+    if dynamic_enabled and enable_topo_check:
+        engine.check_src_dst_consistency(src_ranks, dst_ranks, "neighbor_allreduce")
+
+    works, gathered, keep_alive = _post_neighbor_exchange(
+        tensor, src_ranks, dst_ranks, dst_weights, dst_weighting_enabled
+    )
+    weights = [src_weights[r] for r in src_ranks]
+    output = torch.empty_like(tensor)
+
+    def finalize():
+        return hip_ext.weighted_combine(output, tensor, self_weight, gathered, weights)
+
+    return engine.submit(
+        engine.auto_name("neighbor.allreduce", name),
+        works,
+        finalize,
+        tensor.device,
+        keep_alive=keep_alive + [output],
+    )
+
+
+def neighbor_allreduce(
+    tensor: torch.Tensor,
+    *,
+    self_weight: Optional[float] = None,
+    src_weights: Optional[Dict[int, float]] = None,
+    dst_weights: Optional[Union[Dict[int, float], List[int]]] = None,
+    enable_topo_check: bool = True,
+    name: Optional[str] = None,
+) -> torch.Tensor:
+    """Weighted average of the tensor with the (in-)neighbors' tensors:
+    ``out = self_weight * x_self + sum_j src_weights[j] * (dst_weight_j *) x_j``.
+
+    With no weight arguments the static virtual topology supplies uniform
+    (or, if ``bf.set_topology(..., is_weighted=True)``, matrix) weights.
+    Passing ``self_weight``/``src_weights``/``dst_weights`` selects dynamic
+    per-call neighbors (the one-peer schedules of
+    ``bluefog_amd.parallel.topology``). The input is not modified.
+    """
+    if (self_weight is None) != (src_weights is None):
+        raise ValueError(
+            "Arguments self_weight and src_weights have to be presented at the same time"
+        )
+    handle = neighbor_allreduce_nonblocking(
+        tensor,
+        self_weight=self_weight,
+        src_weights=src_weights,
+        dst_weights=dst_weights,
+        enable_topo_check=enable_topo_check,
+        name=name,
+    )
+    return engine.synchronize(handle)
+
+
+def neighbor_allreduce_nonblocking(
+    tensor: torch.Tensor,
+    *,
+    self_weight: Optional[float] = None,
+    src_weights: Optional[Dict[int, float]] = None,
+    dst_weights: Optional[Union[Dict[int, float], List[int]]] = None,
+    enable_topo_check: bool = True,
+    name: Optional[str] = None,
+) -> int:
+    if (self_weight is None) != (src_weights is None):
+        raise ValueError(
+            "Arguments self_weight and src_weights have to be presented at the same time"
+        )
+    return _neighbor_allreduce_nonblocking_impl(
+        tensor, self_weight, src_weights, dst_weights, enable_topo_check, name
+    )
+
+
+# ---------------------------------------------------------------------------
+# neighbor_allgather
+# ---------------------------------------------------------------------------
+
+
+def neighbor_allgather(
+    tensor: torch.Tensor,
+    *,
+    src_ranks: Optional[List[int]] = None,
+    dst_ranks: Optional[List[int]] = None,
+    enable_topo_check: bool = True,
+    name: Optional[str] = None,
+) -> torch.Tensor:
+    """Concatenate (dim 0) the tensors of all in-neighbors, in
+    ``src_ranks`` order (or ``bf.in_neighbor_ranks()`` order for the static
+    topology). First dims may differ across ranks."""
+    handle = neighbor_allgather_nonblocking(
+        tensor,
+        src_ranks=src_ranks,
+        dst_ranks=dst_ranks,
+        enable_topo_check=enable_topo_check,
+        name=name,
+    )
+    return engine.synchronize(handle)
+
+
+def neighbor_allgather_nonblocking(
+    tensor: torch.Tensor,
+    *,
+    src_ranks: Optional[List[int]] = None,
+    dst_ranks: Optional[List[int]] = None,
+    enable_topo_check: bool = True,
+    name: Optional[str] = None,
+) -> int:
+    c = ctx()
+    if (src_ranks is None) != (dst_ranks is None):
+        raise ValueError(
+            "Argument dst_ranks and src_ranks have to be presented at the same time"
+        )
+    dynamic = src_ranks is not None
+    if not dynamic:
+        src_ranks = c.in_neighbor_ranks()
+        dst_ranks = c.out_neighbor_ranks()
+    elif enable_topo_check:
+        engine.check_src_dst_consistency(src_ranks, dst_ranks, "neighbor_allgather")
+
+    tensor = tensor.detach()
+    if not tensor.is_contiguous():
+        tensor = tensor.contiguous()
+    d0 = tensor.shape[0] if tensor.dim() else 1
+    rest = list(tensor.shape[1:]) if tensor.dim() else []
+    # ragged first dims need a size pre-exchange (blocking, 8 B per edge)
+    src_d0s = engine.exchange_first_dims(d0, src_ranks, dst_ranks)
+
+    recvs = [tensor.new_empty([n] + rest) for n in src_d0s]
+    ops = [
+        dist.P2POp(dist.irecv, r, src) for src, r in zip(src_ranks, recvs) if r.numel()
+    ]
+    ops += [dist.P2POp(dist.isend, tensor, dst) for dst in dst_ranks]
+    works = engine.batch_p2p(ops)
+    output = tensor.new_empty([sum(src_d0s)] + rest)
+
+    def finalize():
+        off = 0
+        for r in recvs:
+            n = r.shape[0]
+            if n:
+                output.narrow(0, off, n).copy_(r)
+            off += n
+        return output
+
+    return engine.submit(
+        engine.auto_name("neighbor.allgather", name),
+        works,
+        finalize,
+        tensor.device,
+        keep_alive=[tensor, output] + recvs,
+    )
+
+
+# ---------------------------------------------------------------------------
+# hierarchical_neighbor_allreduce
+# ---------------------------------------------------------------------------
+
+
+def hierarchical_neighbor_allreduce(
+    tensor: torch.Tensor,
+    *,
+    self_weight: Optional[float] = None,
+    src_machine_weights: Optional[Dict[int, float]] = None,
+    dst_machine_weights: Optional[Union[Dict[int, float], List[int]]] = None,
+    enable_topo_check: bool = False,
+    name: Optional[str] = None,
+) -> torch.Tensor:
+    """Machine-level neighbor averaging: RCCL allreduce inside each node
+    forms a super-node, local-rank-0 exchanges with neighbor machines'
+    leaders, then an intra-node broadcast distributes the result. The output
+    equals ``(self_weight*sum_local + sum_m w_m * sum_local^m) / local_size``
+    (reference: mpi_controller.cc:471-507, nccl_controller.cc:791-857)."""
+    if (self_weight is None) != (src_machine_weights is None):
+        raise ValueError(
+            "Arguments self_weight and src_machine_weights have to be presented "
+            "at the same time"
+        )
+    handle = hierarchical_neighbor_allreduce_nonblocking(
+        tensor,
+        self_weight=self_weight,
+        src_machine_weights=src_machine_weights,
+        dst_machine_weights=dst_machine_weights,
+        enable_topo_check=enable_topo_check,
+        name=name,
+    )
+    return engine.synchronize(handle)
+
+
+def hierarchical_neighbor_allreduce_nonblocking(
+    tensor: torch.Tensor,
+    *,
+    self_weight: Optional[float] = None,
+    src_machine_weights: Optional[Dict[int, float]] = None,
+    dst_machine_weights: Optional[Union[Dict[int, float], List[int]]] = None,
+    enable_topo_check: bool = False,
+    name: Optional[str] = None,
+) -> int:
+    import numpy as np
+
+    c = ctx()
+    assert c.is_homogeneous(), (
+        "hierarchical_neighbor_allreduce should be used under homogeneous "
+        "environment only"
+    )
+    assert c.local_size() > 1, (
+        "If local size is 1, you should use neighbor allreduce directly."
+    )
+    if (self_weight is None) != (src_machine_weights is None):
+        raise ValueError(
+            "Arguments self_weight and src_machine_weights have to be presented "
+            "at the same time"
+        )
+
+    if self_weight is None and src_machine_weights is None and dst_machine_weights is None:
+        topology = c.load_machine_topology()
+        if topology is None:
+            raise RuntimeError(
+                "Machine topology must be set before the use of hierarchical "
+                "neighbor allreduce"
+            )
+        if c.is_machine_topo_weighted():
+            from bluefog_amd.parallel.topology import GetRecvWeights
+
+            self_weight, src_machine_weights = GetRecvWeights(topology, c.machine_rank())
+        else:
+            w = 1.0 / (len(c.in_neighbor_machine_ranks()) + 1)
+            self_weight = w
+            src_machine_weights = {r: w for r in c.in_neighbor_machine_ranks()}
+        dst_machine_weights = {r: 1.0 for r in c.out_neighbor_machine_ranks()}
+        dst_weighting_enabled = False
+    elif self_weight is not None and src_machine_weights is not None and dst_machine_weights is not None:
+        if not isinstance(src_machine_weights, dict):
+            raise ValueError(
+                "Argument src_machine_weights has to be a dictionary map from the "
+                "(in-)neighbor rank to the weights."
+            )
+        if not isinstance(self_weight, float):
+            raise ValueError("Argument self_weight has to be a float for self rank.")
+        if len(set(dst_machine_weights)) != len(dst_machine_weights):
+            raise ValueError(
+                "Argument dst_machine_weights should only contain the unique ranks."
+            )
+        if isinstance(dst_machine_weights, (list, tuple)):
+            dst_machine_weights = {int(d): 1.0 for d in dst_machine_weights}
+        dst_weighting_enabled = not np.allclose(list(dst_machine_weights.values()), 1.0)
+    else:
+        raise ValueError(
+            "Arguments self_weight, src_machine_weights, dst_machine_weights have "
+            "to be presented at the same time."
+        )
+
+    if enable_topo_check:
+        engine.check_src_dst_consistency(
+            list(src_machine_weights.keys()),
+            list(dst_machine_weights.keys()),
+            "hierarchical_neighbor_allreduce",
+        )
+
+    c.ensure_local_groups()
+    local_size = c.local_size()
+    local_rank = c.local_rank()
+    # machine m's leader is its first rank
+    def leader_of(machine: int) -> int:
+        return machine * local_size
+
+    tensor = tensor.detach()
+    buf = tensor.contiguous().clone()
+    src_machines = list(src_machine_weights.keys())
+    dst_machines = list(dst_machine_weights.keys())
+    weights = [src_machine_weights[m] for m in src_machines]
+    d0 = buf.shape[0] if buf.dim() else 1
+    rest = list(buf.shape[1:]) if buf.dim() else []
+    n_src = len(src_machines)
+    gathered = buf.new_empty([n_src * d0] + rest) if n_src else None
+    output = torch.empty_like(buf)
+    is_cuda = buf.is_cuda
+
+    def run_pipeline():
+        # 1. intra-machine sum over RCCL/gloo
+        w1 = dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=c.local_group, async_op=True)
+        w1.wait()
+        # 2. leaders exchange machine sums with neighbor machines' leaders
+        if local_rank == 0:
+            ops = []
+            for i, m in enumerate(src_machines):
+                ops.append(
+                    dist.P2POp(dist.irecv, gathered.narrow(0, i * d0, d0), leader_of(m))
+                )
+            sends = []
+            for m in dst_machines:
+                w = dst_machine_weights[m]
+                st = buf.mul(w) if (dst_weighting_enabled and w != 1.0) else buf
+                sends.append(st)
+                ops.append(dist.P2POp(dist.isend, st, leader_of(m)))
+            for w_ in engine.batch_p2p(ops):
+                w_.wait()
+        # 3. broadcast the gathered machine sums inside the machine
+        if gathered is not None:
+            w3 = dist.broadcast(
+                gathered, src=leader_of(c.machine_rank()), group=c.local_group, async_op=True
+            )
+            w3.wait()
+        # 4. weighted machine average, then /local_size to undo the local sum
+        hip_ext.weighted_combine(output, buf, self_weight, gathered, weights)
+        hip_ext.scale(output, 1.0 / local_size)
+        return output
+
+    if is_cuda:
+        # all stages are stream-ordered; run them now from the submit hook
+        return engine.submit(
+            engine.auto_name("hierarchical.neighbor.allreduce", name),
+            [],
+            run_pipeline,
+            buf.device,
+            keep_alive=[buf, gathered, output],
+        )
+    # CPU: run lazily at synchronize (gloo waits block the host)
+    return engine.submit(
+        engine.auto_name("hierarchical.neighbor.allreduce", name),
+        [],
+        run_pipeline,
+        buf.device,
+    )
+
+
+# ---------------------------------------------------------------------------
+# pair_gossip
+# ---------------------------------------------------------------------------
+
+
+def pair_gossip(
+    tensor: torch.Tensor,
+    target_rank: int,
+    self_weight: Optional[float] = None,
+    pair_weight: Optional[float] = None,
+    name: Optional[str] = None,
+) -> torch.Tensor:
+    """Two-rank exchange-and-average: ``out = self_weight * x_self +
+    pair_weight * x_target`` (defaults 0.5/0.5). Both ranks must name each
+    other (reference: mpi_controller.cc:747-773)."""
+    handle = pair_gossip_nonblocking(tensor, target_rank, self_weight, pair_weight, name)
+    return engine.synchronize(handle)
+
+
+def pair_gossip_nonblocking(
+    tensor: torch.Tensor,
+    target_rank: int,
+    self_weight: Optional[float] = None,
+    pair_weight: Optional[float] = None,
+    name: Optional[str] = None,
+) -> int:
+    if (self_weight is None) != (pair_weight is None):
+        raise ValueError(
+            "Arguments self_weight and pair_weight have to be presented at the same time"
+        )
+    if self_weight is None:
+        self_weight, pair_weight = 0.5, 0.5
+    tensor = tensor.detach()
+    if not tensor.is_contiguous():
+        tensor = tensor.contiguous()
+    recv = torch.empty_like(tensor)
+    ops = [
+        dist.P2POp(dist.irecv, recv, target_rank),
+        dist.P2POp(dist.isend, tensor, target_rank),
+    ]
+    works = engine.batch_p2p(ops)
+    output = torch.empty_like(tensor)
+
+    def finalize():
+        return hip_ext.weighted_combine(output, tensor, self_weight, recv, [pair_weight])
+
+    return engine.submit(
+        engine.auto_name("pair.gossip", name),
+        works,
+        finalize,
+        tensor.device,
+        keep_alive=[tensor, recv, output],
+    )

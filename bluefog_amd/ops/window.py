@@ -1,0 +1,691 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""One-sided ("window") communication: win_create / win_put / win_get /
+win_accumulate / win_update, distributed mutexes, version counters and the
+push-sum associated-p machinery.
+
+Reference analog: bluefog/torch/mpi_win_ops.cc + nccl_win.cc + the window
+halves of mpi_controller.cc / nccl_controller.cc.
+
+MI355X-native design: the reference *emulates* one-sided GPU
+communication with NCCL pair communicators, an MPI req/ack handshake and a
+passive receive thread. On one MI355X node every GPU pair shares an xGMI
+link and HIP IPC maps peer HBM directly into our address space, so this
+module does **true RMA**: `win_put` launches a kernel whose stores land in
+the destination GPU's buffer with zero destination involvement
+(window_ipc.py). Each window allocates one contiguous per-in-neighbor
+buffer block, so `win_update`'s weighted average is a single
+:func:`~bluefog_amd.ops.hip_ext.weighted_combine` kernel.
+
+Control plane (mutex, versions, associated-p scalars) rides the TCP store —
+replacing MPI_Fetch_and_op spin locks and version windows
+(mpi_controller.cc:1284-1392, 1594-1663).
+
+Every one-sided op executes on a single-worker thread pool (the analog of
+the reference's finalizer pool) so nonblocking puts/gets retire in
+submission order without blocking training.
+"""
+
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from contextlib import contextmanager
+from typing import Dict, List, Optional
+
+import torch
+
+from bluefog_amd.ops import hip_ext
+from bluefog_amd.ops.context import ctx
+from bluefog_amd.ops.handles import handle_manager
+from bluefog_amd.utils.env import win_on_gpu
+from bluefog_amd.utils.logging import get_logger
+from bluefog_amd.utils.timeline import timeline
+
+logger = get_logger()
+
+__all__ = [
+    "win_create",
+    "win_free",
+    "win_update",
+    "win_update_then_collect",
+    "win_put_nonblocking",
+    "win_put",
+    "win_get_nonblocking",
+    "win_get",
+    "win_accumulate_nonblocking",
+    "win_accumulate",
+    "win_poll",
+    "win_wait",
+    "win_mutex",
+    "get_win_version",
+    "get_current_created_window_names",
+    "win_associated_p",
+    "turn_on_win_ops_with_associated_p",
+    "turn_off_win_ops_with_associated_p",
+]
+
+_ops_with_associated_p = False
+
+
+def turn_on_win_ops_with_associated_p() -> None:
+    """Make put/accumulate/update also maintain the scalar p of push-sum."""
+    global _ops_with_associated_p
+    _ops_with_associated_p = True
+
+
+def turn_off_win_ops_with_associated_p() -> None:
+    global _ops_with_associated_p
+    _ops_with_associated_p = False
+
+
+class Window:
+    """Per-name window state on this rank."""
+
+    def __init__(
+        self,
+        name: str,
+        tensor: torch.Tensor,
+        in_ranks: List[int],
+        out_ranks: List[int],
+        zero_init: bool,
+    ):
+        self.name = name
+        self.self_tensor = tensor  # the user's tensor; win_update writes it
+        self.in_ranks = list(in_ranks)
+        self.out_ranks = list(out_ranks)
+        self.device = tensor.device
+        self.lock = threading.RLock()
+        shape = list(tensor.shape) if tensor.dim() else [1]
+        # one contiguous block, one same-shaped slot per in-neighbor, so
+        # win_update is a single weighted_combine kernel over it
+        self.block = tensor.new_zeros([max(len(in_ranks), 1)] + shape)
+        if not zero_init:
+            for i in range(len(in_ranks)):
+                self.block[i].copy_(tensor.reshape(shape))
+        self.associated_p = 1.0
+        # transport attachments (set by the registry)
+        self.ipc = None  # window_ipc.IpcWindowPeers
+        self.cpu_addrs = None  # {rank: (host, port)} for the TCP fallback
+
+    def neighbor_buffer(self, src: int) -> torch.Tensor:
+        i = self.in_ranks.index(src)
+        return self.block[i].reshape(self.self_tensor.shape)
+
+    def slot_index_for(self, in_ranks: List[int], origin: int) -> int:
+        return in_ranks.index(origin)
+
+
+class WindowRegistry:
+    def __init__(self):
+        self._windows: Dict[str, Window] = {}
+        self._lock = threading.Lock()
+        self._server = None  # window_cpu.WindowServer
+        self._client = None  # window_cpu.WindowClient
+        self._executor: Optional[ThreadPoolExecutor] = None
+
+    # -- lifecycle ---------------------------------------------------------
+    def names(self) -> List[str]:
+        with self._lock:
+            return sorted(self._windows.keys())
+
+    def get(self, name: str) -> Window:
+        with self._lock:
+            win = self._windows.get(name)
+        if win is None:
+            raise ValueError(f"bluefog_amd: no window registered under name {name!r}")
+        return win
+
+    def executor(self) -> ThreadPoolExecutor:
+        # one worker => one-sided ops retire in submission order (the
+        # reference's finalizer pool with BLUEFOG_NUM_FINALIZER_THREADS=1)
+        if self._executor is None:
+            self._executor = ThreadPoolExecutor(
+                max_workers=1, thread_name_prefix="bf-win"
+            )
+        return self._executor
+
+    def ensure_cpu_transport(self):
+        from bluefog_amd.ops import window_cpu
+
+        c = ctx()
+        if self._server is None:
+            self._server = window_cpu.WindowServer(self)
+            c.store.set(
+                f"winsrv/{c.rank()}", f"{self._server.host}:{self._server.port}".encode()
+            )
+        if self._client is None:
+            c.store.wait([f"winsrv/{r}" for r in range(c.size())])
+            import socket as _socket
+
+            my_host = _socket.gethostname()
+            addrs = {}
+            for r in range(c.size()):
+                host, port = c.store.get(f"winsrv/{r}").decode().rsplit(":", 1)
+                if host == my_host:
+                    host = "127.0.0.1"
+                addrs[r] = (host, int(port))
+            self._client = window_cpu.WindowClient(addrs)
+        return self._client
+
+    def register(self, win: Window) -> None:
+        with self._lock:
+            if win.name in self._windows:
+                raise ValueError(f"window {win.name!r} already exists")
+            self._windows[win.name] = win
+
+    def free(self, name: str) -> None:
+        with self._lock:
+            win = self._windows.pop(name, None)
+        if win is not None and win.ipc is not None:
+            win.ipc.close()
+
+    def free_all(self) -> None:
+        for n in self.names():
+            self.free(n)
+        if self._server is not None:
+            self._server.shutdown()
+            self._server = None
+        if self._executor is not None:
+            self._executor.shutdown(wait=False)
+            self._executor = None
+
+
+_registry = WindowRegistry()
+
+
+def registry() -> WindowRegistry:
+    if ctx().window_registry is None:
+        ctx().window_registry = _registry
+    return _registry
+
+
+# ---------------------------------------------------------------------------
+# store-key helpers: versions, associated-p
+# ---------------------------------------------------------------------------
+
+
+def _ver_key(name: str, owner: int, nbr: int) -> str:
+    return f"win/{name}/ver/{owner}/{nbr}"
+
+
+def _ack_key(name: str, owner: int, nbr: int) -> str:
+    return f"win/{name}/ack/{owner}/{nbr}"
+
+
+def _p_slot_key(name: str, owner: int, origin: int) -> str:
+    return f"win/{name}/p/{owner}/{origin}"
+
+
+def _p_self_key(name: str, rank: int) -> str:
+    return f"win/{name}/selfp/{rank}"
+
+
+def _p_slot_get(store, name, owner, origin) -> float:
+    try:
+        return float(store.get(_p_slot_key(name, owner, origin)).decode())
+    except Exception:
+        return 0.0
+
+
+def _p_slot_set(store, name, owner, origin, value: float) -> None:
+    store.set(_p_slot_key(name, owner, origin), repr(float(value)).encode())
+
+
+def _p_self_get(store, name, rank) -> float:
+    try:
+        return float(store.get(_p_self_key(name, rank)).decode())
+    except Exception:
+        return 1.0
+
+
+def _p_self_set(store, name, rank, value: float) -> None:
+    store.set(_p_self_key(name, rank), repr(float(value)).encode())
+
+
+def _mutex_name(name: str, rank: int) -> str:
+    return f"win.{name}.rank{rank}"
+
+
+# ---------------------------------------------------------------------------
+# public API
+# ---------------------------------------------------------------------------
+
+
+def win_create(tensor: torch.Tensor, name: str, zero_init: bool = False) -> bool:
+    """Create a window for ``tensor`` under ``name`` (collective, blocking).
+
+    Allocates one buffer per in-neighbor (initialized to the tensor's value
+    unless ``zero_init``); on CUDA the buffers and the tensor are IPC-shared
+    so out-neighbors can write them directly over xGMI."""
+    c = ctx()
+    reg = registry()
+    if not tensor.is_contiguous():
+        raise ValueError("win_create requires a contiguous tensor")
+    use_gpu_path = tensor.is_cuda and win_on_gpu()
+    win = Window(
+        name,
+        tensor,
+        c.in_neighbor_ranks(),
+        c.out_neighbor_ranks(),
+        zero_init,
+    )
+    reg.register(win)
+    _p_self_set(c.store, name, c.rank(), 1.0)
+    win.associated_p = 1.0
+
+    if use_gpu_path:
+        from bluefog_amd.ops import window_ipc
+
+        try:
+            win.ipc = window_ipc.IpcWindowPeers(win, c)
+        except Exception as e:
+            logger.warning(
+                "bluefog_amd: HIP IPC window path unavailable (%s); "
+                "falling back to the TCP window server",
+                e,
+            )
+            win.ipc = None
+    if win.ipc is None:
+        reg.ensure_cpu_transport()
+    # all ranks must finish registration before any one-sided traffic
+    c.store.barrier(f"win_create/{name}")
+    return True
+
+
+def win_free(name: Optional[str] = None) -> bool:
+    c = ctx()
+    reg = registry()
+    names = [name] if name else reg.names()
+    if reg._executor is not None:
+        reg.executor().shutdown(wait=True)
+        reg._executor = None
+    c.store.barrier(f"win_free/{'all' if not name else name}")
+    for n in names:
+        reg.free(n)
+    c.store.barrier(f"win_free_done/{'all' if not name else name}")
+    return True
+
+
+def get_current_created_window_names() -> List[str]:
+    return registry().names()
+
+
+def get_win_version(name: str) -> Dict[int, int]:
+    """{in-neighbor: number of unsynced updates to its buffer} (0 = the
+    buffer content has been seen by win_update)."""
+    c = ctx()
+    win = registry().get(name)
+    me = c.rank()
+    out = {}
+    for nbr in win.in_ranks:
+        puts = c.store.counter(_ver_key(name, me, nbr))
+        try:
+            acked = int(c.store.get(_ack_key(name, me, nbr)).decode())
+        except Exception:
+            acked = 0
+        out[nbr] = puts - acked
+    return out
+
+
+def win_associated_p(name: str) -> float:
+    return registry().get(name).associated_p
+
+
+@contextmanager
+def win_mutex(name: str, for_self: bool = False, ranks: Optional[List[int]] = None):
+    """Acquire the distributed window mutex of the given ranks (default: all
+    out-neighbors), or of self with ``for_self=True``."""
+    c = ctx()
+    registry().get(name)  # raise early on unknown window
+    _ranks = c.out_neighbor_ranks() if ranks is None else list(ranks)
+    targets = [c.rank()] if for_self else sorted(_ranks)
+    for r in targets:
+        c.store.mutex_acquire(_mutex_name(name, r))
+    try:
+        yield
+    finally:
+        for r in reversed(targets):
+            c.store.mutex_release(_mutex_name(name, r))
+
+
+# ---------------------------------------------------------------------------
+# win_update
+# ---------------------------------------------------------------------------
+
+
+def win_update(
+    name: str,
+    self_weight: Optional[float] = None,
+    neighbor_weights: Optional[Dict[int, float]] = None,
+    reset: bool = False,
+    clone: bool = False,
+    require_mutex: bool = False,
+) -> torch.Tensor:
+    """Weighted-average the window tensor with its neighbor buffers, in
+    place (unless ``clone``): ``t = self_weight*t + sum_j w_j * buf_j``."""
+    c = ctx()
+    win = registry().get(name)
+    tensor = win.self_tensor
+    if clone:
+        tensor = tensor.clone()
+
+    if neighbor_weights is not None and self_weight is not None:
+        if not isinstance(neighbor_weights, dict):
+            raise ValueError(
+                "Argument neighbor_weights has to be a dictionary map from the "
+                "(in-)neighbor rank to the weights."
+            )
+        if not isinstance(self_weight, float):
+            raise ValueError("Argument self_weight has to be a float for self rank.")
+        if not set(neighbor_weights.keys()).issubset(set(c.in_neighbor_ranks())):
+            raise ValueError(
+                "The key of weights should only contain the ranks that belong to "
+                " in-neighbors and self rank."
+            )
+    elif neighbor_weights is None and self_weight is None:
+        if c.is_topo_weighted():
+            from bluefog_amd.parallel.topology import GetRecvWeights
+
+            self_weight, neighbor_weights = GetRecvWeights(c.load_topology(), c.rank())
+        else:
+            w = 1.0 / (len(c.in_neighbor_ranks()) + 1)
+            self_weight = w
+            neighbor_weights = {r: w for r in c.in_neighbor_ranks()}
+    else:
+        raise ValueError(
+            "Arguments self_weight and neighbor_weights have to be presented at "
+            "the same time"
+        )
+
+    me = c.rank()
+    timeline().start_activity(name, "WIN_UPDATE")
+    if require_mutex:
+        c.store.mutex_acquire(_mutex_name(name, me))
+    try:
+        with win.lock:
+            nbrs = [r for r in win.in_ranks if r in neighbor_weights]
+            weights = [neighbor_weights[r] for r in nbrs]
+            tensors = [win.neighbor_buffer(r) for r in nbrs]
+            hip_ext.weighted_combine_list(tensor, tensor.clone(), self_weight, tensors, weights)
+            if _ops_with_associated_p:
+                p = win.associated_p * self_weight
+                for r, w in zip(nbrs, weights):
+                    p += w * _p_slot_get(c.store, name, me, r)
+                win.associated_p = p
+                _p_self_set(c.store, name, me, p)
+            if reset:
+                reset_ranks = nbrs if neighbor_weights else win.in_ranks
+                for r in reset_ranks:
+                    win.neighbor_buffer(r).zero_()
+                    if _ops_with_associated_p:
+                        _p_slot_set(c.store, name, me, r, 0.0)
+            # mark buffers as seen
+            for r in win.in_ranks:
+                puts = c.store.counter(_ver_key(name, me, r))
+                c.store.set(_ack_key(name, me, r), str(puts).encode())
+    finally:
+        if require_mutex:
+            c.store.mutex_release(_mutex_name(name, me))
+        timeline().end_activity(name)
+    if clone is False and tensor.data_ptr() != win.self_tensor.data_ptr():
+        win.self_tensor.copy_(tensor)
+    return tensor
+
+
+def win_update_then_collect(name: str, require_mutex: bool = True) -> torch.Tensor:
+    """win_update with unit weights + reset — accumulate-then-clear
+    (reference mpi_ops.py:1064-1079)."""
+    c = ctx()
+    neighbor_weights = {r: 1.0 for r in c.in_neighbor_ranks()}
+    return win_update(name, 1.0, neighbor_weights, reset=True, require_mutex=require_mutex)
+
+
+# ---------------------------------------------------------------------------
+# one-sided data-plane ops (run on the window worker thread)
+# ---------------------------------------------------------------------------
+
+
+def _submit_win_op(name: str, op_label: str, job) -> int:
+    h = handle_manager().allocate(f"{op_label}.{name}.{_op_seq()}")
+    timeline().start_activity(name, op_label.upper())
+    h.future = registry().executor().submit(job)
+    return h.id
+
+
+_op_counter = [0]
+_op_counter_lock = threading.Lock()
+
+
+def _op_seq() -> int:
+    with _op_counter_lock:
+        _op_counter[0] += 1
+        return _op_counter[0]
+
+
+def _capture_ready_event(tensor: torch.Tensor):
+    if tensor.is_cuda:
+        ev = torch.cuda.Event()
+        ev.record(torch.cuda.current_stream())
+        return ev
+    return None
+
+
+def win_put_nonblocking(
+    tensor: torch.Tensor,
+    name: str,
+    self_weight: Optional[float] = None,
+    dst_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> int:
+    """Write ``tensor * dst_weights[dst]`` into each destination's buffer
+    for this rank — over xGMI on GPU, with no destination involvement. After
+    the sends, ``tensor`` is scaled in place by ``self_weight``."""
+    c = ctx()
+    win = registry().get(name)
+    dst_weights = (
+        {r: 1.0 for r in c.out_neighbor_ranks()} if dst_weights is None else dst_weights
+    )
+    if self_weight is None:
+        self_weight = 1.0
+    if not set(dst_weights.keys()).issubset(set(c.out_neighbor_ranks())):
+        raise ValueError(
+            "The key of dst_weights should only contain ranks that "
+            " belong to out-neighbors (self-rank is not allowed)."
+        )
+    me = c.rank()
+    ready = _capture_ready_event(tensor)
+
+    def job():
+        if ready is not None:
+            torch.cuda.set_device(win.device)
+            ready.synchronize()
+        my_p = win.associated_p
+        for dst in sorted(dst_weights.keys()):
+            w = dst_weights[dst]
+            if require_mutex:
+                c.store.mutex_acquire(_mutex_name(name, dst))
+            try:
+                if win.ipc is not None:
+                    win.ipc.put(dst, tensor, w)
+                else:
+                    registry().ensure_cpu_transport().put(
+                        dst, name, me, tensor if w == 1.0 else tensor.mul(w)
+                    )
+                if _ops_with_associated_p:
+                    _p_slot_set(c.store, name, dst, me, my_p * w)
+                c.store.add(_ver_key(name, dst, me), 1)
+            finally:
+                if require_mutex:
+                    c.store.mutex_release(_mutex_name(name, dst))
+        # post-send in-place self scaling (reference DoWinPut callback)
+        if self_weight != 1.0:
+            with win.lock:
+                hip_ext.scale(tensor, self_weight)
+                if tensor.is_cuda:
+                    torch.cuda.current_stream().synchronize()
+        if _ops_with_associated_p:
+            win.associated_p = my_p * self_weight
+            _p_self_set(c.store, name, me, win.associated_p)
+        return True
+
+    return _submit_win_op(name, "win.put", job)
+
+
+def win_put(
+    tensor: torch.Tensor,
+    name: str,
+    self_weight: Optional[float] = None,
+    dst_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> bool:
+    return win_wait(win_put_nonblocking(tensor, name, self_weight, dst_weights, require_mutex))
+
+
+def win_accumulate_nonblocking(
+    tensor: torch.Tensor,
+    name: str,
+    self_weight: Optional[float] = None,
+    dst_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> int:
+    """Add ``tensor * dst_weights[dst]`` into each destination's buffer for
+    this rank (SUM only, like the reference). Single-writer-per-slot makes
+    the remote read-modify-write race-free without atomics."""
+    c = ctx()
+    win = registry().get(name)
+    dst_weights = (
+        {r: 1.0 for r in c.out_neighbor_ranks()} if dst_weights is None else dst_weights
+    )
+    if self_weight is None:
+        self_weight = 1.0
+    if not set(dst_weights.keys()).issubset(set(c.out_neighbor_ranks())):
+        raise ValueError(
+            "The key of dst_weights should only contain ranks that "
+            " belong to out-neighbors (self-rank is not allowed)."
+        )
+    me = c.rank()
+    ready = _capture_ready_event(tensor)
+
+    def job():
+        if ready is not None:
+            torch.cuda.set_device(win.device)
+            ready.synchronize()
+        my_p = win.associated_p
+        for dst in sorted(dst_weights.keys()):
+            w = dst_weights[dst]
+            if require_mutex:
+                c.store.mutex_acquire(_mutex_name(name, dst))
+            try:
+                if win.ipc is not None:
+                    win.ipc.accumulate(dst, tensor, w)
+                else:
+                    registry().ensure_cpu_transport().accum(
+                        dst, name, me, tensor if w == 1.0 else tensor.mul(w)
+                    )
+                if _ops_with_associated_p:
+                    old = _p_slot_get(c.store, name, dst, me)
+                    _p_slot_set(c.store, name, dst, me, old + my_p * w)
+                c.store.add(_ver_key(name, dst, me), 1)
+            finally:
+                if require_mutex:
+                    c.store.mutex_release(_mutex_name(name, dst))
+        if self_weight != 1.0:
+            with win.lock:
+                hip_ext.scale(tensor, self_weight)
+                if tensor.is_cuda:
+                    torch.cuda.current_stream().synchronize()
+        if _ops_with_associated_p:
+            win.associated_p = my_p * self_weight
+            _p_self_set(c.store, name, me, win.associated_p)
+        return True
+
+    return _submit_win_op(name, "win.accumulate", job)
+
+
+def win_accumulate(
+    tensor: torch.Tensor,
+    name: str,
+    self_weight: Optional[float] = None,
+    dst_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> bool:
+    return win_wait(
+        win_accumulate_nonblocking(tensor, name, self_weight, dst_weights, require_mutex)
+    )
+
+
+def win_get_nonblocking(
+    name: str,
+    src_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> int:
+    """Fetch each source's current window tensor (scaled by
+    ``src_weights[src]``) into this rank's buffer for that source; a later
+    win_update folds them in."""
+    c = ctx()
+    win = registry().get(name)
+    src_weights = (
+        {r: 1.0 for r in c.in_neighbor_ranks()} if src_weights is None else src_weights
+    )
+    if not set(src_weights.keys()).issubset(set(c.in_neighbor_ranks())):
+        raise ValueError(
+            "The key of src_weights should only contain ranks that "
+            " belong to in-neighbors."
+        )
+    me = c.rank()
+
+    def job():
+        if win.device.type == "cuda":
+            torch.cuda.set_device(win.device)
+        for src in sorted(src_weights.keys()):
+            w = src_weights[src]
+            if require_mutex:
+                c.store.mutex_acquire(_mutex_name(name, src))
+            try:
+                with win.lock:
+                    buf = win.neighbor_buffer(src)
+                    if win.ipc is not None:
+                        win.ipc.get(src, buf, w)
+                    else:
+                        data = registry().ensure_cpu_transport().get(src, name)
+                        buf.copy_(data.to(buf.device))
+                        if w != 1.0:
+                            buf.mul_(w)
+                if _ops_with_associated_p:
+                    src_p = _p_self_get(c.store, name, src)
+                    _p_slot_set(c.store, name, me, src, src_p * w)
+                c.store.add(_ver_key(name, me, src), 1)
+            finally:
+                if require_mutex:
+                    c.store.mutex_release(_mutex_name(name, src))
+        if win.device.type == "cuda":
+            torch.cuda.current_stream().synchronize()
+        return True
+
+    return _submit_win_op(name, "win.get", job)
+
+
+def win_get(
+    name: str,
+    src_weights: Optional[Dict[int, float]] = None,
+    require_mutex: bool = False,
+) -> bool:
+    return win_wait(win_get_nonblocking(name, src_weights, require_mutex))
+
+
+# deprecated aliases kept for API parity
+neighbor_win_put = win_put
+neighbor_win_put_nonblocking = win_put_nonblocking
+neighbor_win_get = win_get
+neighbor_win_get_nonblocking = win_get_nonblocking
+neighbor_win_accumulate = win_accumulate
+neighbor_win_accumulate_nonblocking = win_accumulate_nonblocking
+
+
+def win_poll(handle: int) -> bool:
+    return handle_manager().poll(handle)
+
+
+def win_wait(handle: int) -> bool:
+    h = handle_manager().get(handle)
+    h.synchronize()
+    handle_manager().release(handle)
+    return True

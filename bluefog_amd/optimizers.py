@@ -1,0 +1,1091 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Distributed optimizer wrappers — the training-loop layer.
+
+Reference analog: bluefog/torch/optimizers.py (all five wrapper families,
+same factory names, same hook points, same ``synchronize``/
+``skip_synchronize``/``num_steps_per_communication`` semantics).
+
+MI355X-native difference: where the reference fires one
+nonblocking op per parameter and relies on runtime request fusion
+(operations.cc:943-1020), the neighbor/allreduce wrappers here flatten the
+model's parameters into a few persistent contiguous *buckets* at
+construction (params become views into the bucket storage, so there is no
+per-iteration pack/unpack at all) and communicate one bucket per RCCL
+group — a handful of ~64 MB xGMI transfers per step, overlapped with
+forward+backward, finished by one fused weighted-average HIP kernel each.
+"""
+
+import os
+import warnings
+from contextlib import contextmanager
+from enum import Enum
+from typing import Dict, List, Optional
+
+import torch
+
+from bluefog_amd.utils.env import fusion_threshold_bytes
+from bluefog_amd.utils.logging import get_logger
+
+logger = get_logger()
+
+
+class CommunicationType(Enum):
+    neighbor_allreduce = "neighbor.allreduce"
+    hierarchical_neighbor_allreduce = "hierarchical.neighbor.allreduce"
+    allreduce = "allreduce"
+    empty = "empty"
+
+
+_warning_message_num_step_per_communication = (
+    "Unexpected behavior: num_steps_per_communication is smaller than the "
+    "number of forward passes between two optimizer steps. Communication "
+    "was triggered more often than configured."
+)
+_warning_message_backward_pass_per_step = (
+    "Unexpected behavior: backward_passes_per_step is smaller than the "
+    "number of backward passes between two optimizer steps."
+)
+
+
+def _bf():
+    import bluefog_amd as bf
+
+    return bf
+
+
+def _named_leaf_module(module, parent_name=None):
+    """Yield (name, submodule) for every leaf module."""
+    if next(module.named_children(), None) is None:
+        yield (parent_name, module)
+    for name, ch in module.named_children():
+        full = name if parent_name is None else f"{parent_name}.{name}"
+        yield from _named_leaf_module(ch, full)
+
+
+def _find_duplicates(lst):
+    seen, dup = set(), set()
+    for el in lst:
+        if el in seen:
+            dup.add(el)
+        seen.add(el)
+    return dup
+
+
+def _check_named_parameters(optimizer, model):
+    """Validate model(s) vs optimizer params; returns (named_parameters,
+    models) like the reference (optimizers.py:72-109)."""
+    if isinstance(model, torch.nn.Module):
+        models = [model]
+    elif isinstance(model, (list, tuple)) and all(
+        isinstance(m, torch.nn.Module) for m in model
+    ):
+        models = list(model)
+    else:
+        raise TypeError("model must be a torch.nn.Module or a list of them")
+    named_parameters = []
+    for m in models:
+        named_parameters.extend(m.named_parameters())
+    all_param_ids = {
+        id(p) for group in optimizer.param_groups for p in group["params"]
+    }
+    named_param_ids = {id(p) for _, p in named_parameters}
+    unnamed = all_param_ids - named_param_ids
+    if unnamed:
+        raise ValueError(
+            "named_parameters was specified, but one or more model parameters "
+            "were not named. To use the distributed optimizer every optimizer "
+            "parameter must come from the given model(s)."
+        )
+    dups = _find_duplicates([k for k, _ in named_parameters])
+    if dups:
+        raise ValueError(
+            f"Parameter names in the model(s) must be unique; duplicates: {sorted(dups)}"
+        )
+    return named_parameters, models
+
+
+# ---------------------------------------------------------------------------
+# flat parameter buckets
+# ---------------------------------------------------------------------------
+
+
+class _FlatBuckets:
+    """Flatten parameters into persistent contiguous per-(device,dtype)
+    buckets and rebind each ``p.data`` to a view of its bucket."""
+
+    def __init__(self, named_params, bucket_bytes: int):
+        self.buckets = []  # each: dict(flat=Tensor, params=[Parameter], name=str)
+        groups: Dict[tuple, list] = {}
+        seen = set()
+        for name, p in named_params:
+            if not p.requires_grad or id(p) in seen:
+                continue
+            seen.add(id(p))
+            groups.setdefault((str(p.device), p.dtype), []).append((name, p))
+        bidx = 0
+        for (_, dtype), plist in groups.items():
+            cur: List = []
+            cur_bytes = 0
+            esize = torch.empty(0, dtype=dtype).element_size()
+            for name, p in plist:
+                cur.append((name, p))
+                cur_bytes += p.numel() * esize
+                if cur_bytes >= bucket_bytes:
+                    self._seal(cur, bidx)
+                    bidx += 1
+                    cur, cur_bytes = [], 0
+            if cur:
+                self._seal(cur, bidx)
+                bidx += 1
+
+    def _seal(self, named_plist, bidx: int) -> None:
+        total = sum(p.numel() for _, p in named_plist)
+        first = named_plist[0][1]
+        flat = torch.empty(total, dtype=first.dtype, device=first.device)
+        off = 0
+        with torch.no_grad():
+            for _, p in named_plist:
+                n = p.numel()
+                seg = flat.narrow(0, off, n)
+                seg.copy_(p.data.reshape(-1))
+                p.data = seg.view(p.shape)
+                off += n
+        self.buckets.append(
+            {
+                "flat": flat,
+                "params": [p for _, p in named_plist],
+                "name": f"bucket.{bidx}",
+            }
+        )
+
+    def __len__(self):
+        return len(self.buckets)
+
+
+# ---------------------------------------------------------------------------
+# 1. gradient allreduce (Horovod-style synchronous DP)
+# ---------------------------------------------------------------------------
+
+
+class _DistributedOptimizer(torch.optim.Optimizer):
+    """Synchronous gradient averaging over all ranks (reference
+    optimizers.py:166-294): per-parameter nonblocking allreduce fired from a
+    post-grad-accumulation hook, synchronized at step()."""
+
+    def __init__(self, params, model, backward_passes_per_step=1):
+        super(self.__class__, self).__init__(params)
+        named_parameters, models = _check_named_parameters(self, model)
+        self._models = models
+        self._parameter_names = {v: k for k, v in sorted(named_parameters)}
+        self._handles = {}
+        self._synchronized = False
+        self._should_synchronize = True
+        self._error_encountered = False
+        self._backward_passes_per_step = backward_passes_per_step
+        self._allreduce_delay = {
+            v: self._backward_passes_per_step for _, v in sorted(named_parameters)
+        }
+        self._timeline_hook_handles = []
+        self._use_timeline = False
+        if _bf().size() > 1:
+            self._register_hooks()
+
+    def _bluefog_base_step(self, closure=None):
+        return super(self.__class__, self).step(closure)
+
+    def _register_hooks(self):
+        for param_group in self.param_groups:
+            for p in param_group["params"]:
+                if p.requires_grad:
+                    p.register_post_accumulate_grad_hook(self._make_hook())
+
+    def _make_hook(self):
+        def hook(p):
+            if self._allreduce_delay[p] <= 0:
+                if not self._error_encountered:
+                    warnings.warn(_warning_message_backward_pass_per_step)
+                    self._error_encountered = True
+            self._allreduce_delay[p] -= 1
+            if self._allreduce_delay[p] == 0:
+                handle = _bf().allreduce_nonblocking_(
+                    p.grad, average=True, name=self._parameter_names.get(p)
+                )
+                self._handles[p] = handle
+
+        return hook
+
+    def turn_on_timeline(self):
+        self._use_timeline = True
+
+    def turn_off_timeline(self):
+        self._use_timeline = False
+
+    def synchronize(self):
+        with torch.no_grad():
+            for p, handle in self._handles.items():
+                _bf().synchronize(handle)
+                self._allreduce_delay[p] = self._backward_passes_per_step
+        self._handles.clear()
+        self._synchronized = True
+
+    @contextmanager
+    def skip_synchronize(self):
+        self._should_synchronize = False
+        try:
+            yield
+        finally:
+            self._should_synchronize = True
+
+    def step(self, closure=None):
+        if self._should_synchronize:
+            if self._synchronized:
+                warnings.warn(
+                    "optimizer.step() called without optimizer.skip_synchronize() "
+                    "context after optimizer.synchronize(). Consider the "
+                    "skip_synchronize() context."
+                )
+            self.synchronize()
+        self._synchronized = False
+        return super(self.__class__, self).step(closure)
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self._handles:
+            raise AssertionError(
+                "optimizer.zero_grad() was called after loss.backward() but "
+                "before optimizer.step() or optimizer.synchronize()."
+            )
+        return super(self.__class__, self).zero_grad(set_to_none)
+
+
+# ---------------------------------------------------------------------------
+# 2. adapt-with-combine (CTA / consensus): the flagship family
+# ---------------------------------------------------------------------------
+
+
+class _DistributedReduceOptimizer(torch.optim.Optimizer):
+    """Parameter averaging launched from a model forward hook, overlapped
+    with forward+backward; step() = synchronize then base optimizer step
+    (reference optimizers.py:297-482). Uses flat buckets (class docstring)."""
+
+    def __init__(self, params, model, communication_type, num_steps_per_communication=1):
+        super(self.__class__, self).__init__(params)
+        named_parameters, models = _check_named_parameters(self, model)
+        # dynamic-topology knobs, read at every hook firing
+        self.self_weight = None
+        self.src_weights = None
+        self.dst_weights = None
+        self.src_machine_weights = None
+        self.dst_machine_weights = None
+        self.enable_topo_check = False
+
+        self._models = models
+        self._parameter_names = {v: k for k, v in sorted(named_parameters)}
+        self._name_parameters = {k: v for k, v in sorted(named_parameters)}
+        self._handles = {}  # bucket-name -> handle
+        self._synchronized = False
+        self._should_synchronize = True
+        self._error_encountered = False
+        self._num_steps_per_communication = num_steps_per_communication
+        assert isinstance(communication_type, CommunicationType)
+        self._communication_type = communication_type
+        self._reduce_delay = num_steps_per_communication
+        self._timeline_hook_handles = []
+        self._use_timeline = False
+
+        self._buckets = _FlatBuckets(named_parameters, fusion_threshold_bytes())
+        if os.getenv("BLUEFOG_TIMELINE"):
+            self.turn_on_timeline()
+        if _bf().size() > 1:
+            self._register_hooks()
+
+    def _bluefog_base_step(self, closure=None):
+        return super(self.__class__, self).step(closure)
+
+    def _register_hooks(self):
+        for model in self._models:
+            model.register_forward_hook(self._make_hook())
+
+    def _make_hook(self):
+        def hook(model, *unused):
+            if not model.training:
+                return
+            if self._handles:
+                return  # this step's communication is already in flight
+            if self._reduce_delay <= 0:
+                if not self._error_encountered:
+                    warnings.warn(_warning_message_num_step_per_communication)
+                    self._error_encountered = True
+            self._reduce_delay -= 1
+            if self._reduce_delay == 0:
+                self._launch_communication()
+
+        return hook
+
+    def _launch_communication(self):
+        bf = _bf()
+        for b in self._buckets.buckets:
+            name = b["name"]
+            if self._communication_type == CommunicationType.allreduce:
+                h = bf.allreduce_nonblocking(b["flat"], average=True, name=name)
+            elif self._communication_type == CommunicationType.neighbor_allreduce:
+                h = bf.neighbor_allreduce_nonblocking(
+                    b["flat"],
+                    name=name,
+                    self_weight=self.self_weight,
+                    src_weights=self.src_weights,
+                    dst_weights=self.dst_weights,
+                    enable_topo_check=self.enable_topo_check,
+                )
+            elif self._communication_type == CommunicationType.hierarchical_neighbor_allreduce:
+                h = bf.hierarchical_neighbor_allreduce_nonblocking(
+                    b["flat"],
+                    name=name,
+                    self_weight=self.self_weight,
+                    src_machine_weights=self.src_machine_weights,
+                    dst_machine_weights=self.dst_machine_weights,
+                    enable_topo_check=self.enable_topo_check,
+                )
+            elif self._communication_type == CommunicationType.empty:
+                h = None
+            else:
+                raise ValueError("Unsupported CommunicationType encountered.")
+            self._handles[name] = (b, h)
+
+    @property
+    def communication_type(self):
+        return self._communication_type
+
+    @communication_type.setter
+    def communication_type(self, value):
+        assert isinstance(value, CommunicationType)
+        self._communication_type = value
+
+    def turn_on_timeline(self):
+        self._use_timeline = True
+
+    def turn_off_timeline(self):
+        self._use_timeline = False
+
+    def synchronize(self):
+        bf = _bf()
+        with torch.no_grad():
+            for name, (b, handle) in self._handles.items():
+                if handle is not None:
+                    output = bf.synchronize(handle)
+                    b["flat"].copy_(output)
+            self._reduce_delay = self._num_steps_per_communication
+        self._handles.clear()
+        self._synchronized = True
+
+    @contextmanager
+    def skip_synchronize(self):
+        self._should_synchronize = False
+        try:
+            yield
+        finally:
+            self._should_synchronize = True
+
+    def step(self, closure=None):
+        if self._should_synchronize:
+            if self._synchronized:
+                warnings.warn(
+                    "optimizer.step() called without optimizer.skip_synchronize() "
+                    "context after optimizer.synchronize(). Consider the "
+                    "skip_synchronize() context."
+                )
+            self.synchronize()
+        self._synchronized = False
+        return super(self.__class__, self).step(closure)
+
+
+# ---------------------------------------------------------------------------
+# 3. adapt-then-combine
+# ---------------------------------------------------------------------------
+
+
+class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
+    """ATC: each parameter's gradient hook first applies the optimizer step
+    for that parameter, then launches neighbor averaging of the *updated*
+    weights — communication overlaps the rest of backward (reference
+    optimizers.py:485-841)."""
+
+    def __init__(self, params, model, communication_type, backward_passes_per_step=1):
+        super(self.__class__, self).__init__(params)
+        named_parameters, models = _check_named_parameters(self, model)
+        self.self_weight = None
+        self.src_weights = None
+        self.dst_weights = None
+        self.src_machine_weights = None
+        self.dst_machine_weights = None
+        self.enable_topo_check = False
+
+        self._models = models
+        self._parameter_names = {v: k for k, v in sorted(named_parameters)}
+        self._handles = {}
+        self._synchronized = False
+        self._should_synchronize = True
+        self._error_encountered = False
+        self._backward_passes_per_step = backward_passes_per_step
+        assert isinstance(communication_type, CommunicationType)
+        self._communication_type = communication_type
+        self._reduce_delay = {
+            v: backward_passes_per_step for _, v in sorted(named_parameters)
+        }
+        self._step_func = None
+        self._timeline_hook_handles = []
+        self._use_timeline = False
+        if _bf().size() > 1:
+            self._register_hooks()
+
+    def _bluefog_base_step(self, closure=None):
+        return super(self.__class__, self).step(closure)
+
+    @property
+    def communication_type(self):
+        return self._communication_type
+
+    @communication_type.setter
+    def communication_type(self, value):
+        assert isinstance(value, CommunicationType)
+        self._communication_type = value
+
+    def register_step_function(self, step_func):
+        """step_func(parameter, gradient, param_group) applies the update for
+        one parameter; replaces the built-in sgd/adam/... steps."""
+        self._step_func = step_func
+
+    def _register_hooks(self):
+        for param_group in self.param_groups:
+            for p in param_group["params"]:
+                if p.requires_grad:
+                    p.register_post_accumulate_grad_hook(
+                        self._make_hook(param_group)
+                    )
+
+    def _make_hook(self, param_group):
+        def hook(p):
+            if self._reduce_delay[p] <= 0:
+                if not self._error_encountered:
+                    warnings.warn(_warning_message_num_step_per_communication)
+                    self._error_encountered = True
+            self._reduce_delay[p] -= 1
+            if self._reduce_delay[p] == 0:
+                with torch.no_grad():
+                    if self._step_func is not None:
+                        self._step_func(p, p.grad, param_group)
+                    else:
+                        self._builtin_step(p, p.grad, param_group)
+                    self._handles[p] = self._launch_comm_for(p)
+
+        return hook
+
+    # -- parameter-wise optimizer steps (reference optimizers.py:601-760) --
+    def _builtin_step(self, p, grad, group):
+        if {"momentum", "dampening", "nesterov"}.issubset(group.keys()):
+            self._sgd_step(p, grad, group)
+        elif {"betas", "eps", "amsgrad"}.issubset(group.keys()):
+            self._adam_step(p, grad, group)
+        elif {"alpha", "eps", "centered"}.issubset(group.keys()):
+            self._rmsprop_step(p, grad, group)
+        elif {"lr_decay", "eps"}.issubset(group.keys()):
+            self._adagrad_step(p, grad, group)
+        elif {"rho", "eps"}.issubset(group.keys()):
+            self._adadelta_step(p, grad, group)
+        else:
+            raise ValueError(
+                "Cannot infer the optimizer type for the ATC step; register a "
+                "custom step with register_step_function()."
+            )
+
+    def _sgd_step(self, p, grad, group):
+        wd, momentum, dampening = group["weight_decay"], group["momentum"], group["dampening"]
+        nesterov, lr = group["nesterov"], group["lr"]
+        d_p = grad
+        if wd != 0:
+            d_p = d_p.add(p.data, alpha=wd)
+        if momentum != 0:
+            state = self.state[p]
+            buf = state.get("momentum_buffer")
+            if buf is None:
+                buf = torch.clone(d_p).detach()
+                state["momentum_buffer"] = buf
+            else:
+                buf.mul_(momentum).add_(d_p, alpha=1 - dampening)
+            d_p = d_p.add(buf, alpha=momentum) if nesterov else buf
+        p.data.add_(d_p, alpha=-lr)
+
+    def _adam_step(self, p, grad, group):
+        beta1, beta2 = group["betas"]
+        eps, lr, wd = group["eps"], group["lr"], group["weight_decay"]
+        amsgrad = group["amsgrad"]
+        state = self.state[p]
+        if len(state) == 0:
+            state["step"] = 0
+            state["exp_avg"] = torch.zeros_like(p.data)
+            state["exp_avg_sq"] = torch.zeros_like(p.data)
+            if amsgrad:
+                state["max_exp_avg_sq"] = torch.zeros_like(p.data)
+        if wd != 0:
+            grad = grad.add(p.data, alpha=wd)
+        state["step"] += 1
+        t = state["step"]
+        exp_avg, exp_avg_sq = state["exp_avg"], state["exp_avg_sq"]
+        exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
+        exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+        bias1 = 1 - beta1 ** t
+        bias2 = 1 - beta2 ** t
+        if amsgrad:
+            torch.maximum(state["max_exp_avg_sq"], exp_avg_sq, out=state["max_exp_avg_sq"])
+            denom = (state["max_exp_avg_sq"] / bias2).sqrt_().add_(eps)
+        else:
+            denom = (exp_avg_sq / bias2).sqrt_().add_(eps)
+        p.data.addcdiv_(exp_avg, denom, value=-lr / bias1)
+
+    def _rmsprop_step(self, p, grad, group):
+        alpha, eps, lr = group["alpha"], group["eps"], group["lr"]
+        wd, momentum, centered = group["weight_decay"], group["momentum"], group["centered"]
+        state = self.state[p]
+        if len(state) == 0:
+            state["step"] = 0
+            state["square_avg"] = torch.zeros_like(p.data)
+            if momentum > 0:
+                state["momentum_buffer"] = torch.zeros_like(p.data)
+            if centered:
+                state["grad_avg"] = torch.zeros_like(p.data)
+        if wd != 0:
+            grad = grad.add(p.data, alpha=wd)
+        state["step"] += 1
+        square_avg = state["square_avg"]
+        square_avg.mul_(alpha).addcmul_(grad, grad, value=1 - alpha)
+        if centered:
+            grad_avg = state["grad_avg"]
+            grad_avg.mul_(alpha).add_(grad, alpha=1 - alpha)
+            avg = square_avg.addcmul(grad_avg, grad_avg, value=-1).sqrt_().add_(eps)
+        else:
+            avg = square_avg.sqrt().add_(eps)
+        if momentum > 0:
+            buf = state["momentum_buffer"]
+            buf.mul_(momentum).addcdiv_(grad, avg)
+            p.data.add_(buf, alpha=-lr)
+        else:
+            p.data.addcdiv_(grad, avg, value=-lr)
+
+    def _adagrad_step(self, p, grad, group):
+        lr, lr_decay, wd, eps = group["lr"], group["lr_decay"], group["weight_decay"], group["eps"]
+        state = self.state[p]
+        if len(state) == 0:
+            state["step"] = 0
+            state["sum"] = torch.zeros_like(p.data)
+        if wd != 0:
+            grad = grad.add(p.data, alpha=wd)
+        state["step"] += 1
+        clr = lr / (1 + (state["step"] - 1) * lr_decay)
+        state["sum"].addcmul_(grad, grad, value=1)
+        std = state["sum"].sqrt().add_(eps)
+        p.data.addcdiv_(grad, std, value=-clr)
+
+    def _adadelta_step(self, p, grad, group):
+        rho, eps, lr, wd = group["rho"], group["eps"], group["lr"], group["weight_decay"]
+        state = self.state[p]
+        if len(state) == 0:
+            state["step"] = 0
+            state["square_avg"] = torch.zeros_like(p.data)
+            state["acc_delta"] = torch.zeros_like(p.data)
+        if wd != 0:
+            grad = grad.add(p.data, alpha=wd)
+        state["step"] += 1
+        square_avg, acc_delta = state["square_avg"], state["acc_delta"]
+        square_avg.mul_(rho).addcmul_(grad, grad, value=1 - rho)
+        std = square_avg.add(eps).sqrt_()
+        delta = acc_delta.add(eps).sqrt_().div_(std).mul_(grad)
+        p.data.add_(delta, alpha=-lr)
+        acc_delta.mul_(rho).addcmul_(delta, delta, value=1 - rho)
+
+    # -- communication -----------------------------------------------------
+    def _launch_comm_for(self, p):
+        bf = _bf()
+        name = self._parameter_names.get(p)
+        if self._communication_type == CommunicationType.allreduce:
+            return bf.allreduce_nonblocking(p.data, average=True, name=name)
+        if self._communication_type == CommunicationType.neighbor_allreduce:
+            return bf.neighbor_allreduce_nonblocking(
+                p.data,
+                name=name,
+                self_weight=self.self_weight,
+                src_weights=self.src_weights,
+                dst_weights=self.dst_weights,
+                enable_topo_check=self.enable_topo_check,
+            )
+        if self._communication_type == CommunicationType.hierarchical_neighbor_allreduce:
+            return bf.hierarchical_neighbor_allreduce_nonblocking(
+                p.data,
+                name=name,
+                self_weight=self.self_weight,
+                src_machine_weights=self.src_machine_weights,
+                dst_machine_weights=self.dst_machine_weights,
+                enable_topo_check=self.enable_topo_check,
+            )
+        if self._communication_type == CommunicationType.empty:
+            return None
+        raise ValueError("Unsupported CommunicationType encountered.")
+
+    def turn_on_timeline(self):
+        self._use_timeline = True
+
+    def turn_off_timeline(self):
+        self._use_timeline = False
+
+    def synchronize(self):
+        bf = _bf()
+        with torch.no_grad():
+            for p, handle in self._handles.items():
+                if handle is not None:
+                    output = bf.synchronize(handle)
+                    p.set_(output)
+                self._reduce_delay[p] = self._backward_passes_per_step
+        self._handles.clear()
+        self._synchronized = True
+
+    @contextmanager
+    def skip_synchronize(self):
+        self._should_synchronize = False
+        try:
+            yield
+        finally:
+            self._should_synchronize = True
+
+    def step(self, closure=None):
+        # the hooks already applied the parameter update; step() only
+        # synchronizes the in-flight neighbor averaging
+        if self._should_synchronize:
+            if self._synchronized:
+                warnings.warn(
+                    "optimizer.step() called without optimizer.skip_synchronize() "
+                    "context after optimizer.synchronize()."
+                )
+            self.synchronize()
+        self._synchronized = False
+        if closure is not None:
+            with torch.enable_grad():
+                return closure()
+        return None
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self._handles:
+            raise AssertionError(
+                "optimizer.zero_grad() was called after loss.backward() but "
+                "before optimizer.step() or optimizer.synchronize()."
+            )
+        return super(self.__class__, self).zero_grad(set_to_none)
+
+
+# ---------------------------------------------------------------------------
+# 4. window (async gossip) optimizers
+# ---------------------------------------------------------------------------
+
+
+class _DistributedWinOptimizer(torch.optim.Optimizer):
+    """Async gossip through one-sided windows: push (win_put) or pull
+    (win_get) neighbor parameters each forward, fold them in with
+    win_update at step() (reference optimizers.py:844-1023)."""
+
+    def __init__(self, params, model, num_steps_per_communication, window_prefix, pull_style):
+        super(self.__class__, self).__init__(params)
+        if pull_style:
+            self.src_weights = None
+        else:
+            self.dst_weights = None
+        self.force_barrier = False
+        self.window_prefix = window_prefix + "." if window_prefix is not None else ""
+
+        named_parameters, models = _check_named_parameters(self, model)
+        self._models = models
+        self._pull_style = pull_style
+        self._parameter_names = {
+            v: self.window_prefix + k for k, v in sorted(named_parameters)
+        }
+        self._handles = {}
+        self._synchronized = False
+        self._should_synchronize = True
+        self._error_encountered = False
+        self._num_steps_per_communication = num_steps_per_communication
+        self._delay = {
+            v: num_steps_per_communication for _, v in sorted(named_parameters)
+        }
+        self._timeline_hook_handles = []
+        self._use_timeline = False
+        if _bf().size() > 1:
+            self._register_window()
+            self._register_hooks()
+
+    def _bluefog_base_step(self, closure=None):
+        return super(self.__class__, self).step(closure)
+
+    def __del__(self):
+        try:
+            self.unregister_window()
+        except Exception:
+            pass
+
+    def _register_hooks(self):
+        for model in self._models:
+            hook = self._make_get_hook() if self._pull_style else self._make_put_hook()
+            model.register_forward_hook(hook)
+
+    def _fire(self, p, name):
+        bf = _bf()
+        if self._pull_style:
+            return bf.win_get_nonblocking(
+                name=name, src_weights=self.src_weights, require_mutex=True
+            )
+        return bf.win_put_nonblocking(
+            tensor=p.data, name=name, dst_weights=self.dst_weights, require_mutex=False
+        )
+
+    def _make_hook_impl(self):
+        def hook(model, *unused):
+            if not model.training:
+                return
+            for p in (
+                q for group in self.param_groups for q in group["params"] if q.requires_grad
+            ):
+                if p in self._handles:
+                    continue
+                if self._delay[p] <= 0:
+                    if not self._error_encountered:
+                        warnings.warn(_warning_message_num_step_per_communication)
+                        self._error_encountered = True
+                self._delay[p] -= 1
+                if self._delay[p] == 0:
+                    self._handles[p] = self._fire(p, self._parameter_names.get(p))
+
+        return hook
+
+    def _make_put_hook(self):
+        return self._make_hook_impl()
+
+    def _make_get_hook(self):
+        return self._make_hook_impl()
+
+    def _register_window(self):
+        bf = _bf()
+        for param_group in self.param_groups:
+            for p in param_group["params"]:
+                name = self._parameter_names.get(p)
+                if name is None:
+                    raise KeyError("Cannot find parameter in _parameter_names")
+                if not bf.win_create(p.data, name):
+                    raise ValueError(f"Cannot allocate window for parameter {name}")
+
+    def unregister_window(self):
+        bf = _bf()
+        if bf.size() <= 1:
+            return
+        for param_group in self.param_groups:
+            for p in param_group["params"]:
+                name = self._parameter_names.get(p)
+                if name in bf.get_current_created_window_names():
+                    bf.win_free(name)
+
+    def turn_on_timeline(self):
+        self._use_timeline = True
+
+    def turn_off_timeline(self):
+        self._use_timeline = False
+
+    @contextmanager
+    def skip_synchronize(self):
+        self._should_synchronize = False
+        try:
+            yield
+        finally:
+            self._should_synchronize = True
+
+    def synchronize(self):
+        bf = _bf()
+        with torch.no_grad():
+            for p, handle in self._handles.items():
+                bf.win_wait(handle)
+                name = self._parameter_names.get(p)
+                self._delay[p] = self._num_steps_per_communication
+                p.set_(bf.win_update(name=name, require_mutex=True))
+        self._handles.clear()
+        self._synchronized = True
+
+    def step(self, closure=None):
+        if self.force_barrier:
+            _bf().barrier()
+        if self._should_synchronize:
+            if self._synchronized:
+                warnings.warn(
+                    "optimizer.step() called without optimizer.skip_synchronize() "
+                    "context after optimizer.synchronize()."
+                )
+            self.synchronize()
+        self._synchronized = False
+        return super(self.__class__, self).step(closure)
+
+
+class _DistributedPushSumOptimizer(torch.optim.Optimizer):
+    """Push-sum gossip with weight correction: windows carry the flattened
+    parameter plus a trailing scalar p; accumulates with column-stochastic
+    weights 1/(outdegree+1); sync divides by the accumulated p (reference
+    optimizers.py:1026-1177)."""
+
+    def __init__(self, params, model, num_steps_per_communication):
+        super(self.__class__, self).__init__(params)
+        bf = _bf()
+        outdegree = len(bf.out_neighbor_ranks())
+        self.dst_weights = {
+            rank: 1.0 / (outdegree + 1) for rank in bf.out_neighbor_ranks()
+        }
+        self.self_weight = 1.0 / (outdegree + 1)
+        self.force_barrier = True
+
+        named_parameters, models = _check_named_parameters(self, model)
+        self._models = models
+        self._parameter_names = {v: k for k, v in sorted(named_parameters)}
+        self._handles = {}
+        self._named_ps_weights = {}
+        self._named_extension_parameters = {}
+        self._synchronized = False
+        self._should_synchronize = True
+        self._error_encountered = False
+        self._num_steps_per_communication = num_steps_per_communication
+        self._delay = {
+            v: num_steps_per_communication for _, v in sorted(named_parameters)
+        }
+        self._timeline_hook_handles = []
+        self._use_timeline = False
+        if bf.size() > 1:
+            self._register_window()
+            self._register_hooks()
+
+    def _bluefog_base_step(self, closure=None):
+        return super(self.__class__, self).step(closure)
+
+    @torch.no_grad()
+    def _register_window(self):
+        bf = _bf()
+        for param_group in self.param_groups:
+            for p in param_group["params"]:
+                name = self._parameter_names.get(p)
+                if name is None:
+                    raise KeyError("Cannot find parameter in _parameter_names")
+                ps_weights = torch.Tensor([1.0]).to(p.data.dtype).to(p.data.device)
+                self._named_ps_weights[name] = ps_weights
+                extended = torch.cat((p.data.view(-1), ps_weights), 0)
+                self._named_extension_parameters[name] = extended
+                if not bf.win_create(extended, name, zero_init=True):
+                    raise ValueError(f"Cannot allocate window for parameter {name}")
+
+    def _register_hooks(self):
+        for model in self._models:
+            model.register_forward_hook(self._make_hook())
+
+    def _make_hook(self):
+        def hook(model, *unused):
+            if not model.training:
+                return
+            bf = _bf()
+            for p in (
+                q for group in self.param_groups for q in group["params"] if q.requires_grad
+            ):
+                if p in self._handles:
+                    continue
+                if self._delay[p] <= 0:
+                    if not self._error_encountered:
+                        warnings.warn(_warning_message_num_step_per_communication)
+                        self._error_encountered = True
+                self._delay[p] -= 1
+                if self._delay[p] == 0:
+                    name = self._parameter_names.get(p)
+                    with torch.no_grad():
+                        ext = self._named_extension_parameters[name]
+                        ext[:-1].copy_(p.data.view(-1))
+                        ext[-1] = self._named_ps_weights[name].item()
+                    self._handles[p] = bf.win_accumulate_nonblocking(
+                        tensor=ext,
+                        name=name,
+                        dst_weights=self.dst_weights,
+                        require_mutex=True,
+                    )
+
+        return hook
+
+    def turn_on_timeline(self):
+        self._use_timeline = True
+
+    def turn_off_timeline(self):
+        self._use_timeline = False
+
+    @contextmanager
+    def skip_synchronize(self):
+        self._should_synchronize = False
+        try:
+            yield
+        finally:
+            self._should_synchronize = True
+
+    def synchronize(self):
+        bf = _bf()
+        with torch.no_grad():
+            for p, handle in self._handles.items():
+                bf.win_wait(handle)
+                name = self._parameter_names.get(p)
+                self._delay[p] = self._num_steps_per_communication
+                ext = self._named_extension_parameters[name]
+                # keep self's share, fold in neighbors' accumulations
+                ext.mul_(self.self_weight)
+                ext = bf.win_update_then_collect(name=name)
+                self._named_ps_weights[name].fill_(ext[-1].item())
+                corrected = (ext[:-1] / ext[-1]).reshape(p.shape)
+                p.set_(corrected)
+        self._handles.clear()
+        self._synchronized = True
+
+    def step(self, closure=None):
+        if self.force_barrier:
+            _bf().barrier()
+        if self._should_synchronize:
+            if self._synchronized:
+                warnings.warn(
+                    "optimizer.step() called without optimizer.skip_synchronize() "
+                    "context after optimizer.synchronize()."
+                )
+            self.synchronize()
+        self._synchronized = False
+        return super(self.__class__, self).step(closure)
+
+
+# ---------------------------------------------------------------------------
+# factories (reference optimizers.py:1180-1554)
+# ---------------------------------------------------------------------------
+
+
+def _wrap(optimizer, cls_impl, *args):
+    cls = type(optimizer.__class__.__name__, (optimizer.__class__,), dict(cls_impl.__dict__))
+    return cls(optimizer.param_groups, *args)
+
+
+def DistributedGradientAllreduceOptimizer(optimizer, model, backward_passes_per_step=1):
+    """Synchronous gradient-averaging DP (Horovod-equivalent)."""
+    return _wrap(optimizer, _DistributedOptimizer, model, backward_passes_per_step)
+
+
+def DistributedAdaptWithCombineOptimizer(
+    optimizer,
+    model,
+    communication_type=CommunicationType.neighbor_allreduce,
+    num_steps_per_communication=1,
+):
+    """CTA/consensus: neighbor-average the parameters (overlapped with
+    fwd+bwd), then apply the local gradient."""
+    return _wrap(
+        optimizer,
+        _DistributedReduceOptimizer,
+        model,
+        communication_type,
+        num_steps_per_communication,
+    )
+
+
+def DistributedAdaptThenCombineOptimizer(
+    optimizer,
+    model,
+    communication_type=CommunicationType.neighbor_allreduce,
+    backward_passes_per_step=1,
+):
+    """ATC: per-parameter optimizer step inside the gradient hook, then
+    neighbor-average the updated weights."""
+    return _wrap(
+        optimizer,
+        _DistributedAdaptThenCombineOptimizer,
+        model,
+        communication_type,
+        backward_passes_per_step,
+    )
+
+
+def DistributedAllreduceOptimizer(optimizer, model, num_steps_per_communication=1):
+    """Deprecated alias: AdaptWithCombine over global allreduce."""
+    warnings.warn(
+        "DistributedAllreduceOptimizer is deprecated; use "
+        "DistributedAdaptWithCombineOptimizer(communication_type="
+        "CommunicationType.allreduce)",
+        DeprecationWarning,
+    )
+    return _wrap(
+        optimizer,
+        _DistributedReduceOptimizer,
+        model,
+        CommunicationType.allreduce,
+        num_steps_per_communication,
+    )
+
+
+def DistributedNeighborAllreduceOptimizer(optimizer, model, num_steps_per_communication=1):
+    """Deprecated alias: AdaptWithCombine over neighbor_allreduce."""
+    warnings.warn(
+        "DistributedNeighborAllreduceOptimizer is deprecated; use "
+        "DistributedAdaptWithCombineOptimizer",
+        DeprecationWarning,
+    )
+    return _wrap(
+        optimizer,
+        _DistributedReduceOptimizer,
+        model,
+        CommunicationType.neighbor_allreduce,
+        num_steps_per_communication,
+    )
+
+
+def DistributedHierarchicalNeighborAllreduceOptimizer(
+    optimizer, model, num_steps_per_communication=1
+):
+    """Deprecated alias: AdaptWithCombine over hierarchical machine-level
+    neighbor_allreduce."""
+    warnings.warn(
+        "DistributedHierarchicalNeighborAllreduceOptimizer is deprecated; use "
+        "DistributedAdaptWithCombineOptimizer(communication_type="
+        "CommunicationType.hierarchical_neighbor_allreduce)",
+        DeprecationWarning,
+    )
+    return _wrap(
+        optimizer,
+        _DistributedReduceOptimizer,
+        model,
+        CommunicationType.hierarchical_neighbor_allreduce,
+        num_steps_per_communication,
+    )
+
+
+def DistributedWinPutOptimizer(optimizer, model, num_steps_per_communication=1, window_prefix=None):
+    """Async push gossip via one-sided win_put."""
+    return _wrap(
+        optimizer,
+        _DistributedWinOptimizer,
+        model,
+        num_steps_per_communication,
+        window_prefix,
+        False,
+    )
+
+
+def DistributedPullGetOptimizer(optimizer, model, num_steps_per_communication=1):
+    """Async pull gossip via one-sided win_get."""
+    return _wrap(
+        optimizer,
+        _DistributedWinOptimizer,
+        model,
+        num_steps_per_communication,
+        None,
+        True,
+    )
+
+
+def DistributedPushSumOptimizer(optimizer, model, num_steps_per_communication=1):
+    """Push-sum with associated-weight correction over win_accumulate."""
+    return _wrap(
+        optimizer, _DistributedPushSumOptimizer, model, num_steps_per_communication
+    )
