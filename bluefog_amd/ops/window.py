@@ -218,11 +218,14 @@ def _p_self_key(name: str, rank: int) -> str:
     return f"win/{name}/selfp/{rank}"
 
 
+# NB: TCPStore.get BLOCKS until the key exists, so every key read below is
+# pre-initialized at win_create (or guarded by check()).
+
+
 def _p_slot_get(store, name, owner, origin) -> float:
-    try:
-        return float(store.get(_p_slot_key(name, owner, origin)).decode())
-    except Exception:
+    if not store.check([_p_slot_key(name, owner, origin)]):
         return 0.0
+    return float(store.get(_p_slot_key(name, owner, origin)).decode())
 
 
 def _p_slot_set(store, name, owner, origin, value: float) -> None:
@@ -230,10 +233,9 @@ def _p_slot_set(store, name, owner, origin, value: float) -> None:
 
 
 def _p_self_get(store, name, rank) -> float:
-    try:
-        return float(store.get(_p_self_key(name, rank)).decode())
-    except Exception:
+    if not store.check([_p_self_key(name, rank)]):
         return 1.0
+    return float(store.get(_p_self_key(name, rank)).decode())
 
 
 def _p_self_set(store, name, rank, value: float) -> None:
@@ -317,10 +319,7 @@ def get_win_version(name: str) -> Dict[int, int]:
     out = {}
     for nbr in win.in_ranks:
         puts = c.store.counter(_ver_key(name, me, nbr))
-        try:
-            acked = int(c.store.get(_ack_key(name, me, nbr)).decode())
-        except Exception:
-            acked = 0
+        acked = c.store.counter(_ack_key(name, me, nbr))
         out[nbr] = puts - acked
     return out
 
@@ -640,12 +639,15 @@ def win_get_nonblocking(
             if require_mutex:
                 c.store.mutex_acquire(_mutex_name(name, src))
             try:
-                with win.lock:
-                    buf = win.neighbor_buffer(src)
-                    if win.ipc is not None:
-                        win.ipc.get(src, buf, w)
-                    else:
-                        data = registry().ensure_cpu_transport().get(src, name)
+                if win.ipc is not None:
+                    with win.lock:
+                        win.ipc.get(src, win.neighbor_buffer(src), w)
+                else:
+                    # fetch over TCP WITHOUT holding our window lock (our own
+                    # server may need it to answer a symmetric get)
+                    data = registry().ensure_cpu_transport().get(src, name)
+                    with win.lock:
+                        buf = win.neighbor_buffer(src)
                         buf.copy_(data.to(buf.device))
                         if w != 1.0:
                             buf.mul_(w)

@@ -1,0 +1,153 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Context/topology basics over real worlds (reference analog:
+test/torch_basics_test.py)."""
+
+import pytest
+import torch
+
+from tests.util import run_dist
+
+
+def w_rank_size():
+    import bluefog_amd as bf
+
+    bf.init()
+    assert 0 <= bf.rank() < bf.size()
+    assert bf.local_size() >= 1
+    assert not bf.mpi_threads_supported()  # no MPI here by design
+    assert bf.nccl_built()
+
+
+def w_set_load_topologies():
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    size, rank = bf.size(), bf.rank()
+    for builder in (
+        tu.ExponentialTwoGraph,
+        tu.RingGraph,
+        tu.StarGraph,
+        tu.FullyConnectedGraph,
+        tu.MeshGrid2DGraph,
+    ):
+        topo = builder(size)
+        assert bf.set_topology(topo)
+        loaded = bf.load_topology()
+        assert tu.IsTopologyEquivalent(topo, loaded)
+        expected_in = sorted(r for r in topo.predecessors(rank) if r != rank)
+        expected_out = sorted(r for r in topo.successors(rank) if r != rank)
+        assert bf.in_neighbor_ranks() == expected_in
+        assert bf.out_neighbor_ranks() == expected_out
+
+
+def w_default_topology_is_exp2():
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    assert tu.IsTopologyEquivalent(bf.load_topology(), tu.ExponentialGraph(bf.size()))
+
+
+def w_weighted_topology():
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    topo = tu.MeshGrid2DGraph(bf.size())
+    bf.set_topology(topo, is_weighted=True)
+    assert bf.is_topo_weighted()
+    t = torch.ones(4, dtype=torch.float64) * (bf.rank() + 1)
+    out = bf.neighbor_allreduce(t)
+    self_w, nbr_w = tu.GetRecvWeights(topo, bf.rank())
+    expected = self_w * (bf.rank() + 1) + sum(w * (r + 1) for r, w in nbr_w.items())
+    assert torch.allclose(out, torch.full((4,), expected, dtype=torch.float64)), (
+        bf.rank(),
+        out,
+        expected,
+    )
+
+
+def w_infer_round_trip():
+    import bluefog_amd as bf
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    # every rank sends to (rank+1)%size and (rank+2)%size
+    dsts = sorted({(rank + 1) % size, (rank + 2) % size} - {rank})
+    srcs = bf.InferSourceFromDestinationRanks(dsts)
+    expected_srcs = sorted({(rank - 1) % size, (rank - 2) % size} - {rank})
+    assert sorted(srcs) == expected_srcs, (srcs, expected_srcs)
+    back = bf.InferDestinationFromSourceRanks(srcs)
+    assert sorted(back) == dsts, (back, dsts)
+    # with adjacency matrix
+    srcs2, W = bf.InferSourceFromDestinationRanks(dsts, construct_adjacency_matrix=True)
+    assert sorted(srcs2) == expected_srcs
+    assert W.shape == (size, size)
+
+
+def w_suspend_resume():
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.suspend()
+    bf.resume()
+    out = bf.allreduce(torch.ones(2))
+    assert torch.allclose(out, torch.ones(2))
+
+
+def w_skip_negotiate():
+    import bluefog_amd as bf
+
+    bf.init()
+    assert bf.get_skip_negotiate_stage()  # default: program-order contract
+    bf.set_skip_negotiate_stage(False)
+    assert not bf.get_skip_negotiate_stage()
+    bf.set_skip_negotiate_stage(True)
+
+
+def test_rank_size():
+    run_dist(w_rank_size, 2)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_set_load_topologies(ws):
+    run_dist(w_set_load_topologies, ws)
+
+
+def test_default_topology_is_exp2():
+    run_dist(w_default_topology_is_exp2, 4)
+
+
+def test_weighted_topology():
+    run_dist(w_weighted_topology, 4)
+
+
+def test_infer_round_trip():
+    run_dist(w_infer_round_trip, 4)
+
+
+def test_suspend_resume():
+    run_dist(w_suspend_resume, 2)
+
+
+def test_skip_negotiate():
+    run_dist(w_skip_negotiate, 2)
+
+
+def test_single_process_init():
+    # also valid without any env: world of one
+    import subprocess, sys
+
+    code = (
+        "import torch, bluefog_amd as bf\n"
+        "bf.init()\n"
+        "assert bf.size() == 1 and bf.rank() == 0\n"
+        "out = bf.neighbor_allreduce(torch.ones(3))\n"
+        "assert torch.allclose(out, torch.ones(3))\n"
+        "print('OK')\n"
+    )
+    r = subprocess.run(
+        [sys.executable, "-c", code], capture_output=True, text=True, timeout=120
+    )
+    assert r.returncode == 0 and "OK" in r.stdout, r.stderr

@@ -1,0 +1,281 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""One-sided window op tests (reference analog: test/torch_win_ops_test.py):
+lifecycle, update weights, put/get/accumulate exact values, versions, mutex,
+associated-p push-sum consistency. CPU path (TCP window server); the GPU/IPC
+path is covered in test_gpu_ops.py."""
+
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from tests.util import run_dist
+
+
+def _init_ring():
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.set_topology(bf.RingGraph(bf.size()))
+    return bf
+
+
+def w_win_lifecycle():
+    bf = _init_ring()
+    t = torch.ones(4) * bf.rank()
+    assert bf.win_create(t, "w0")
+    assert bf.get_current_created_window_names() == ["w0"]
+    if bf.size() > 2:  # for size 2 Star == Ring, set_topology short-circuits
+        # topology change must be refused while a window exists
+        assert not bf.set_topology(bf.StarGraph(bf.size()))
+    assert bf.win_free("w0")
+    assert bf.get_current_created_window_names() == []
+    assert bf.set_topology(bf.StarGraph(bf.size()))
+
+
+def w_win_update_default():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(6) * rank
+    bf.win_create(t, "wu")
+    # no puts yet: neighbor buffers hold creation-time clones of OUR tensor
+    out = bf.win_update("wu")
+    n = len(bf.in_neighbor_ranks())
+    # buffers were initialized with self value => average == self value
+    assert torch.allclose(out, torch.ones(6) * rank)
+    bf.win_free("wu")
+
+
+def w_win_put_update():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(5) * rank
+    bf.win_create(t, "wp", zero_init=True)
+    bf.barrier()
+    assert bf.win_put(t, "wp")
+    bf.barrier()
+    out = bf.win_update("wp")
+    # ring: in-neighbors put their rank values; buffers were zero before
+    nbrs = bf.in_neighbor_ranks()
+    w = 1.0 / (len(nbrs) + 1)
+    expected = w * rank + sum(w * r for r in nbrs)
+    assert torch.allclose(out, torch.full((5,), expected), atol=1e-6), (
+        rank,
+        out[0].item(),
+        expected,
+    )
+    bf.barrier()
+    bf.win_free("wp")
+
+
+def w_win_put_weighted():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(3, dtype=torch.float64) * (rank + 1)
+    bf.win_create(t, "wpw", zero_init=True)
+    bf.barrier()
+    dsts = {r: 0.5 for r in bf.out_neighbor_ranks()}
+    bf.win_put(t, "wpw", dst_weights=dsts)
+    bf.barrier()
+    nbrs = bf.in_neighbor_ranks()
+    nbr_w = {r: 1.0 for r in nbrs}
+    out = bf.win_update("wpw", 1.0, nbr_w)
+    expected = (rank + 1) + sum(0.5 * (r + 1) for r in nbrs)
+    assert torch.allclose(out, torch.full((3,), expected, dtype=torch.float64)), (
+        rank,
+        out,
+        expected,
+    )
+    bf.barrier()
+    bf.win_free("wpw")
+
+
+def w_win_put_self_weight():
+    bf = _init_ring()
+    rank = bf.rank()
+    t = torch.ones(4) * 8.0
+    bf.win_create(t, "wsw", zero_init=True)
+    bf.barrier()
+    bf.win_put(t, "wsw", self_weight=0.5)
+    bf.barrier()
+    # after put, the local tensor is scaled in place by self_weight
+    assert torch.allclose(t, torch.full((4,), 4.0))
+    bf.barrier()
+    bf.win_free("wsw")
+
+
+def w_win_accumulate():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(4) * (rank + 1)
+    bf.win_create(t, "wa", zero_init=True)
+    bf.barrier()
+    for _ in range(3):
+        bf.win_accumulate(t, "wa")
+    bf.barrier()
+    out = bf.win_update("wa", 1.0, {r: 1.0 for r in bf.in_neighbor_ranks()})
+    expected = (rank + 1) + 3 * sum(r + 1 for r in bf.in_neighbor_ranks())
+    assert torch.allclose(out, torch.full((4,), float(expected))), (rank, out, expected)
+    bf.barrier()
+    bf.win_free("wa")
+
+
+def w_win_get():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(4, dtype=torch.float64) * (rank + 1)
+    bf.win_create(t, "wg", zero_init=True)
+    bf.barrier()
+    assert bf.win_get("wg")
+    bf.barrier()
+    out = bf.win_update("wg", 1.0, {r: 1.0 for r in bf.in_neighbor_ranks()})
+    expected = (rank + 1) + sum(r + 1 for r in bf.in_neighbor_ranks())
+    assert torch.allclose(out, torch.full((4,), float(expected), dtype=torch.float64)), (
+        rank,
+        out,
+        expected,
+    )
+    bf.barrier()
+    bf.win_free("wg")
+
+
+def w_win_version():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(2) * rank
+    bf.win_create(t, "wv", zero_init=True)
+    bf.barrier()
+    ver = bf.get_win_version("wv")
+    assert all(v == 0 for v in ver.values()), ver
+    bf.win_put(t, "wv")
+    bf.win_put(t, "wv")
+    bf.barrier()
+    ver = bf.get_win_version("wv")
+    assert all(v == 2 for v in ver.values()), ver
+    bf.win_update("wv")
+    ver = bf.get_win_version("wv")
+    assert all(v == 0 for v in ver.values()), ver
+    bf.barrier()
+    bf.win_free("wv")
+
+
+def w_win_mutex():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    t = torch.zeros(1)
+    bf.win_create(t, "wm")
+    bf.barrier()
+    # serialize increments of a store counter under the self-rank mutex of
+    # rank 0's window: all ranks contend on the same mutex
+    from bluefog_amd.ops.context import ctx
+
+    store = ctx().store
+    with bf.win_mutex("wm", ranks=[0]):
+        v = store.counter("test/mutex/check")
+        time.sleep(0.02)
+        store.reset_counter("test/mutex/check", v + 1)
+    bf.barrier()
+    assert store.counter("test/mutex/check") == size
+    bf.barrier()
+    bf.win_free("wm")
+
+
+def w_associated_p():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    bf.turn_on_win_ops_with_associated_p()
+    try:
+        t = torch.ones(3) * rank
+        bf.win_create(t, "wap", zero_init=True)
+        assert bf.win_associated_p("wap") == pytest.approx(1.0)
+        bf.barrier()
+        outdeg = len(bf.out_neighbor_ranks())
+        w = 1.0 / (outdeg + 1)
+        bf.win_accumulate(t, "wap", self_weight=w, dst_weights={r: w for r in bf.out_neighbor_ranks()})
+        bf.barrier()
+        bf.win_update_then_collect("wap")
+        # mass conservation: sum of p over ranks stays == size
+        p = torch.tensor([bf.win_associated_p("wap")])
+        total = bf.allreduce(p, average=False)
+        assert total.item() == pytest.approx(size, rel=1e-5), total
+        bf.barrier()
+        bf.win_free("wap")
+    finally:
+        bf.turn_off_win_ops_with_associated_p()
+
+
+def w_push_sum_consistency():
+    """Randomized push-sum over the extended-scalar pattern: after enough
+    rounds every rank's corrected value approaches the global average
+    (reference analog: torch_win_ops_test.py:780-863)."""
+    import bluefog_amd as bf
+
+    bf.init()
+    size, rank = bf.size(), bf.rank()
+    bf.set_topology(bf.ExponentialTwoGraph(size))
+    x = torch.tensor([float(rank)], dtype=torch.float64)
+    ext = torch.cat([x, torch.ones(1, dtype=torch.float64)])
+    bf.win_create(ext, "ps", zero_init=True)
+    bf.barrier()
+    outdeg = len(bf.out_neighbor_ranks())
+    w = 1.0 / (outdeg + 1)
+    for _ in range(40):
+        bf.win_accumulate(
+            ext, "ps", dst_weights={r: w for r in bf.out_neighbor_ranks()},
+            require_mutex=True,
+        )
+        bf.barrier()
+        ext.mul_(w)
+        ext = bf.win_update_then_collect("ps")
+        bf.barrier()
+    avg = ext[0] / ext[1]
+    expected = sum(range(size)) / size
+    assert abs(avg.item() - expected) < 1e-6, (rank, avg.item(), expected)
+    bf.win_free("ps")
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_win_lifecycle(ws):
+    run_dist(w_win_lifecycle, ws)
+
+
+def test_win_update_default():
+    run_dist(w_win_update_default, 4)
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_win_put_update(ws):
+    run_dist(w_win_put_update, ws)
+
+
+def test_win_put_weighted():
+    run_dist(w_win_put_weighted, 4)
+
+
+def test_win_put_self_weight():
+    run_dist(w_win_put_self_weight, 2)
+
+
+def test_win_accumulate():
+    run_dist(w_win_accumulate, 4)
+
+
+def test_win_get():
+    run_dist(w_win_get, 4)
+
+
+def test_win_version():
+    run_dist(w_win_version, 2)
+
+
+def test_win_mutex():
+    run_dist(w_win_mutex, 4)
+
+
+def test_associated_p():
+    run_dist(w_associated_p, 4)
+
+
+def test_push_sum_consistency():
+    run_dist(w_push_sum_consistency, 4, timeout=300)
