@@ -1,0 +1,270 @@
+#!/usr/bin/env python3
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Flagship benchmark — BASELINE.json's headline metric.
+
+ResNet50, batch 64/GPU, synthetic ImageNet-shaped data, random-init
+weights, decentralized data parallelism with dynamic one-peer
+Exponential-2 neighbor_allreduce (the reference's headline configuration:
+examples/pytorch_benchmark.py --dist-optimizer=neighbor_allreduce,
+docs/performance.rst:12-24; reference V100 number: 269.4 img/s/GPU at 95%+
+scaling).
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 run one rank per GPU via torch.distributed.run (reads
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env):
+
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints exactly one JSON line with the whole-job aggregate
+images/sec. Timing: W untimed warmup steps, then exactly K steps bracketed
+by barrier + torch.cuda.synchronize() on both sides; elapsed = MAX over
+ranks.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--gpus", type=int, default=int(os.environ.get("WORLD_SIZE", "1")))
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "resnet101", "resnet18", "bert_base"])
+    p.add_argument("--batch-size", type=int, default=64)
+    p.add_argument("--seq-len", type=int, default=512, help="bert only")
+    p.add_argument(
+        "--dist-optimizer",
+        default="neighbor_allreduce",
+        choices=[
+            "neighbor_allreduce",
+            "hierarchical_neighbor_allreduce",
+            "gradient_allreduce",
+            "allreduce",
+            "win_put",
+            "pushsum",
+            "atc",
+            "local",
+        ],
+    )
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16", "fp16"])
+    p.add_argument("--no-channels-last", action="store_true")
+    p.add_argument("--device", default=None, help="cuda|cpu (default: auto)")
+    p.add_argument("--profile-steps", type=int, default=0,
+                   help="if >0, run only this many timed steps (profiling aid)")
+    return p.parse_args()
+
+
+def build_model_and_data(args, device):
+    if args.model == "bert_base":
+        from bluefog_amd.models import bert_base
+
+        torch.manual_seed(42)
+        model = bert_base().to(device)
+        gen = torch.Generator().manual_seed(1 + int(os.environ.get("RANK", "0")))
+        batches = []
+        for _ in range(4):
+            ids = torch.randint(0, 30522, (args.batch_size, args.seq_len), generator=gen)
+            batches.append((ids.to(device), ids.to(device)))
+
+        def step_fn(model, optimizer, batch):
+            ids, labels = batch
+            optimizer.zero_grad()
+            loss = model(ids, labels=labels)
+            loss.backward()
+            optimizer.step()
+
+        return model, batches, step_fn
+    from bluefog_amd.models import resnet18, resnet50, resnet101
+
+    torch.manual_seed(42)
+    model = {"resnet50": resnet50, "resnet101": resnet101, "resnet18": resnet18}[
+        args.model
+    ]().to(device)
+    if device.type == "cuda" and not args.no_channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    gen = torch.Generator().manual_seed(1 + int(os.environ.get("RANK", "0")))
+    batches = []
+    for _ in range(4):
+        x = torch.randn(args.batch_size, 3, 224, 224, generator=gen)
+        y = torch.randint(0, 1000, (args.batch_size,), generator=gen)
+        if device.type == "cuda" and not args.no_channels_last:
+            x = x.to(memory_format=torch.channels_last)
+        batches.append((x.to(device), y.to(device)))
+    loss_fn = torch.nn.CrossEntropyLoss()
+    amp_dtype = {"bf16": torch.bfloat16, "fp16": torch.float16}.get(args.dtype)
+
+    def step_fn(model, optimizer, batch):
+        x, y = batch
+        optimizer.zero_grad()
+        if amp_dtype is not None:
+            with torch.autocast(device_type=device.type, dtype=amp_dtype):
+                loss = loss_fn(model(x), y)
+        else:
+            loss = loss_fn(model(x), y)
+        loss.backward()
+        optimizer.step()
+
+    return model, batches, step_fn
+
+
+def main():
+    args = parse_args()
+    import bluefog_amd as bf
+
+    bf.init()
+    if args.device:
+        device = torch.device(args.device)
+    else:
+        device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    if device.type == "cuda":
+        torch.backends.cudnn.benchmark = True  # MIOpen find mode
+    n = bf.size()
+    rank = bf.rank()
+
+    topo = bf.ExponentialTwoGraph(n)
+    bf.set_topology(topo)
+    model, batches, step_fn = build_model_and_data(args, device)
+
+    base_opt = torch.optim.SGD(model.parameters(), lr=0.0125 * n, momentum=0.9)
+    dyn_gen = None
+    opt_name = args.dist_optimizer
+    if opt_name == "neighbor_allreduce":
+        optimizer = bf.DistributedAdaptWithCombineOptimizer(
+            base_opt, model=model,
+            communication_type=bf.CommunicationType.neighbor_allreduce,
+        )
+        if n > 1:
+            import bluefog_amd.parallel.topology as tu
+
+            dyn_gen = tu.GetDynamicOnePeerSendRecvRanks(topo, rank)
+    elif opt_name == "hierarchical_neighbor_allreduce":
+        if bf.machine_size() > 1:
+            bf.set_machine_topology(bf.ExponentialTwoGraph(bf.machine_size()))
+            ct = bf.CommunicationType.hierarchical_neighbor_allreduce
+        else:
+            # one machine: machine-level averaging degenerates to a node-local
+            # allreduce, which is exactly CommunicationType.allreduce
+            ct = bf.CommunicationType.allreduce if n > 1 else bf.CommunicationType.empty
+        optimizer = bf.DistributedAdaptWithCombineOptimizer(
+            base_opt, model=model, communication_type=ct
+        )
+    elif opt_name == "gradient_allreduce":
+        optimizer = bf.DistributedGradientAllreduceOptimizer(base_opt, model=model)
+    elif opt_name == "allreduce":
+        optimizer = bf.DistributedAdaptWithCombineOptimizer(
+            base_opt, model=model, communication_type=bf.CommunicationType.allreduce
+        )
+    elif opt_name == "win_put":
+        optimizer = bf.DistributedWinPutOptimizer(base_opt, model=model)
+    elif opt_name == "pushsum":
+        optimizer = bf.DistributedPushSumOptimizer(base_opt, model=model)
+    elif opt_name == "atc":
+        optimizer = bf.DistributedAdaptThenCombineOptimizer(
+            base_opt, model=model,
+            communication_type=bf.CommunicationType.neighbor_allreduce,
+        )
+        if n > 1:
+            import bluefog_amd.parallel.topology as tu
+
+            dyn_gen = tu.GetDynamicOnePeerSendRecvRanks(topo, rank)
+    else:  # local: no communication (scaling upper bound)
+        optimizer = bf.DistributedAdaptWithCombineOptimizer(
+            base_opt, model=model, communication_type=bf.CommunicationType.empty
+        )
+
+    bf.broadcast_parameters(model.state_dict(), root_rank=0)
+
+    def set_dynamic():
+        if dyn_gen is None:
+            return
+        send, recv = next(dyn_gen)
+        w = 1.0 / (len(recv) + 1)
+        optimizer.self_weight = w
+        optimizer.src_weights = {r: w for r in recv}
+        optimizer.dst_weights = send
+        optimizer.enable_topo_check = False
+
+    model.train()
+    steps = args.profile_steps or args.steps
+
+    for i in range(args.warmup):
+        set_dynamic()
+        step_fn(model, optimizer, batches[i % len(batches)])
+
+    bf.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(steps):
+        set_dynamic()
+        step_fn(model, optimizer, batches[i % len(batches)])
+    bf.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    e = torch.tensor([elapsed], dtype=torch.float64)
+    import torch.distributed as dist
+
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)
+    elapsed = float(e.item())
+
+    if args.model == "bert_base":
+        per_step_items = args.batch_size * args.seq_len * n
+        unit = "tokens/s"
+        gb = args.batch_size * n
+    else:
+        per_step_items = args.batch_size * n
+        unit = "images/s"
+        gb = args.batch_size * n
+    value = per_step_items * steps / elapsed
+    # reference headline: 4310.6 img/s on 16 V100 = 269.4 img/s/GPU
+    vs_baseline = (
+        value / (269.4125 * n)
+        if (args.model == "resnet50" and args.batch_size == 64 and unit == "images/s")
+        else None
+    )
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "images/sec/GPU + scaling efficiency, ResNet50 bs=64"
+                    if args.model == "resnet50"
+                    else f"{unit}, {args.model}",
+                    "value": value,
+                    "unit": unit,
+                    "n_gpus": n,
+                    "steps": steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": elapsed / steps * 1000.0,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": vs_baseline,
+                    "dtype": args.dtype,
+                    "data": "synthetic",
+                    "config": {
+                        "model": args.model,
+                        "global_batch": gb,
+                        "seq_len": args.seq_len if args.model == "bert_base" else 224,
+                        "parallelism": f"decentralized-dp{n} "
+                        f"({args.dist_optimizer}, dynamic one-peer exp2)"
+                        if dyn_gen is not None
+                        else f"decentralized-dp{n} ({args.dist_optimizer})",
+                    },
+                }
+            ),
+            flush=True,
+        )
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,176 @@
+// Copyright 2026. Licensed under the Apache License, Version 2.0.
+//
+// pybind11/torch bindings for the bluefog_amd native kernels
+// (reference analog: the pybind layer of bluefog/torch/mpi_ops.cc:572-690,
+// minus the op queueing — scheduling lives in Python over stream-ordered
+// RCCL).
+
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <string>
+#include <vector>
+
+extern "C" {
+hipError_t bf_weighted_combine(void* out, const void* self, float self_w,
+                               const void* gathered, const float* w, int n_nbr,
+                               long numel, int dtype, hipStream_t stream);
+hipError_t bf_scale_put(void* dst, const void* src, float w, long numel,
+                        int dtype, bool accum, hipStream_t stream);
+hipError_t bf_scale_inplace(void* buf, float f, long numel, int dtype,
+                            hipStream_t stream);
+hipError_t bf_combine_sgd(void* p, float self_w, const void* gathered,
+                          const float* w, int n_nbr, const void* grad,
+                          void* mom, float lr, float mu, float wd,
+                          float dampening, int nesterov, long numel, int dtype,
+                          hipStream_t stream);
+hipError_t bf_combine_adam(void* p, float self_w, const void* gathered,
+                           const float* w, int n_nbr, const void* grad,
+                           float* exp_avg, float* exp_avg_sq, float lr,
+                           float beta1, float beta2, float eps, float wd,
+                           float bias1, float bias2, long numel, int dtype,
+                           hipStream_t stream);
+}
+
+namespace {
+
+int dtype_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat:
+      return 0;
+    case at::kDouble:
+      return 1;
+    case at::kHalf:
+      return 2;
+    case at::kBFloat16:
+      return 3;
+    default:
+      TORCH_CHECK(false, "bluefog_amd kernels support f32/f64/f16/bf16, got ",
+                  t.scalar_type());
+  }
+}
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_hip(hipError_t e, const char* what) {
+  TORCH_CHECK(e == hipSuccess, "bluefog_amd ", what, ": ", hipGetErrorString(e));
+}
+
+void weighted_combine(at::Tensor output, at::Tensor self, double self_weight,
+                      at::Tensor gathered, std::vector<double> weights) {
+  TORCH_CHECK(output.is_contiguous() && self.is_contiguous(),
+              "weighted_combine needs contiguous tensors");
+  TORCH_CHECK(output.numel() == self.numel(), "output/self numel mismatch");
+  const int n = static_cast<int>(weights.size());
+  if (n > 0) {
+    TORCH_CHECK(gathered.is_contiguous(), "gathered must be contiguous");
+    TORCH_CHECK(gathered.numel() == static_cast<long>(n) * self.numel(),
+                "gathered must hold one self-shaped slice per weight");
+    TORCH_CHECK(gathered.scalar_type() == self.scalar_type(), "dtype mismatch");
+  }
+  std::vector<float> w(weights.begin(), weights.end());
+  check_hip(bf_weighted_combine(output.data_ptr(), self.data_ptr(),
+                                static_cast<float>(self_weight),
+                                n ? gathered.data_ptr() : self.data_ptr(),
+                                w.data(), n, self.numel(), dtype_code(self),
+                                current_stream()),
+            "weighted_combine");
+}
+
+void scale_put(at::Tensor dst, at::Tensor src, double weight) {
+  TORCH_CHECK(dst.is_contiguous() && src.is_contiguous(), "contiguous only");
+  TORCH_CHECK(dst.numel() == src.numel(), "numel mismatch");
+  TORCH_CHECK(dst.scalar_type() == src.scalar_type(), "dtype mismatch");
+  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(),
+                         static_cast<float>(weight), src.numel(),
+                         dtype_code(src), /*accum=*/false, current_stream()),
+            "scale_put");
+}
+
+void accum_put(at::Tensor dst, at::Tensor src, double weight) {
+  TORCH_CHECK(dst.is_contiguous() && src.is_contiguous(), "contiguous only");
+  TORCH_CHECK(dst.numel() == src.numel(), "numel mismatch");
+  TORCH_CHECK(dst.scalar_type() == src.scalar_type(), "dtype mismatch");
+  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(),
+                         static_cast<float>(weight), src.numel(),
+                         dtype_code(src), /*accum=*/true, current_stream()),
+            "accum_put");
+}
+
+void scale_inplace(at::Tensor buf, double factor) {
+  TORCH_CHECK(buf.is_contiguous(), "contiguous only");
+  check_hip(bf_scale_inplace(buf.data_ptr(), static_cast<float>(factor),
+                             buf.numel(), dtype_code(buf), current_stream()),
+            "scale_inplace");
+}
+
+void weighted_combine_sgd(at::Tensor param, double self_weight,
+                          at::Tensor gathered, std::vector<double> weights,
+                          at::Tensor grad, at::Tensor momentum_buf, double lr,
+                          double momentum, double weight_decay,
+                          double dampening, bool nesterov) {
+  TORCH_CHECK(param.is_contiguous() && grad.is_contiguous(), "contiguous only");
+  TORCH_CHECK(param.numel() == grad.numel(), "param/grad numel mismatch");
+  const int n = static_cast<int>(weights.size());
+  std::vector<float> w(weights.begin(), weights.end());
+  void* mom = nullptr;
+  if (momentum_buf.defined() && momentum_buf.numel() > 0) {
+    TORCH_CHECK(momentum_buf.numel() == param.numel(), "momentum numel");
+    mom = momentum_buf.data_ptr();
+  }
+  check_hip(
+      bf_combine_sgd(param.data_ptr(), static_cast<float>(self_weight),
+                     n ? gathered.data_ptr() : param.data_ptr(), w.data(), n,
+                     grad.data_ptr(), mom, static_cast<float>(lr),
+                     static_cast<float>(momentum),
+                     static_cast<float>(weight_decay),
+                     static_cast<float>(dampening), nesterov ? 1 : 0,
+                     param.numel(), dtype_code(param), current_stream()),
+      "weighted_combine_sgd");
+}
+
+void weighted_combine_adam(at::Tensor param, double self_weight,
+                           at::Tensor gathered, std::vector<double> weights,
+                           at::Tensor grad, at::Tensor exp_avg,
+                           at::Tensor exp_avg_sq, double lr, double beta1,
+                           double beta2, double eps, double weight_decay,
+                           long step) {
+  TORCH_CHECK(param.is_contiguous() && grad.is_contiguous(), "contiguous only");
+  TORCH_CHECK(exp_avg.scalar_type() == at::kFloat &&
+                  exp_avg_sq.scalar_type() == at::kFloat,
+              "adam state must be fp32");
+  const int n = static_cast<int>(weights.size());
+  std::vector<float> w(weights.begin(), weights.end());
+  const float bias1 = 1.0f - std::pow(static_cast<float>(beta1), step);
+  const float bias2 = 1.0f - std::pow(static_cast<float>(beta2), step);
+  check_hip(
+      bf_combine_adam(param.data_ptr(), static_cast<float>(self_weight),
+                      n ? gathered.data_ptr() : param.data_ptr(), w.data(), n,
+                      grad.data_ptr(), exp_avg.data_ptr<float>(),
+                      exp_avg_sq.data_ptr<float>(), static_cast<float>(lr),
+                      static_cast<float>(beta1), static_cast<float>(beta2),
+                      static_cast<float>(eps), static_cast<float>(weight_decay),
+                      bias1, bias2, param.numel(), dtype_code(param),
+                      current_stream()),
+      "weighted_combine_adam");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "bluefog_amd native CDNA4 kernels (gfx950)";
+  m.def("weighted_combine", &weighted_combine,
+        "out = self_w*self + sum_k w[k]*gathered[k]");
+  m.def("scale_put", &scale_put, "dst = w*src (dst may be xGMI peer memory)");
+  m.def("accum_put", &accum_put, "dst += w*src (dst may be xGMI peer memory)");
+  m.def("scale_inplace", &scale_inplace, "buf *= f");
+  m.def("weighted_combine_sgd", &weighted_combine_sgd,
+        "fused neighbor-average + SGD(momentum) over a flat bucket");
+  m.def("weighted_combine_adam", &weighted_combine_adam,
+        "fused neighbor-average + Adam over a flat bucket");
+}

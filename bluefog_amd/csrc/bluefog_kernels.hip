@@ -1,0 +1,512 @@
+// Copyright 2026. Licensed under the Apache License, Version 2.0.
+//
+// Hand-written CDNA4 (gfx950) kernels for bluefog_amd.
+//
+// These replace (a) the reference's only CUDA kernel family — buffer scaling,
+// cuda/cuda_kernels.cu:24-116 — and (b) the chains of torch slice arithmetic
+// the reference runs after communication (mpi_ops.cc:99-164 neighbor
+// averaging, mpi_win_ops.cc:185-279 window averaging, optimizers.py:601-760
+// parameter-wise steps).
+//
+// Every kernel here is an HBM3E-bandwidth-bound elementwise stream (the
+// (k+1)-input weighted average moves k+2 values per output element and does
+// k+1 FMAs — arithmetic intensity << 1 FLOP/byte), so the design targets the
+// ~6.3 TB/s achievable HBM bandwidth, not MFMA:
+//   * 16 bytes per lane per access (dwordx4 loads/stores), grid-stride
+//   * 256-thread workgroups (4 waves of 64), grid sized >> 256 workgroups so
+//     all 8 XCDs fill regardless of the b%8 dispatch pattern
+//   * fp32 accumulation for f16/bf16 inputs (matches the reference's CPU
+//     half upcast, mpi_ops.cc:71-83, and torch's fp32 reference numerics)
+//
+// scale_put / accum_put may target IPC-mapped PEER memory: their stores
+// traverse the xGMI link to the destination GPU (the one-sided win_put /
+// win_accumulate data plane). Peer traffic is not cached by the writing
+// GPU, and completion is published by the host-side version-counter store
+// after a stream synchronize, so plain stores are sufficient.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+#define BF_CHECK_HIP(cmd)                                                     \
+  do {                                                                        \
+    hipError_t e = (cmd);                                                     \
+    if (e != hipSuccess) {                                                    \
+      return e;                                                               \
+    }                                                                         \
+  } while (0)
+
+namespace {
+
+constexpr int kThreads = 256;
+constexpr int kMaxNbrPerLaunch = 16;  // weights ride the kernarg segment
+
+struct WeightsArg {
+  float w[kMaxNbrPerLaunch];
+};
+
+// Conversion traits: torch builds with __HIP_NO_HALF_CONVERSIONS__, so
+// half/bf16 <-> float go through the explicit intrinsics.
+template <typename T>
+struct AccOf {
+  using type = float;
+  static __device__ __forceinline__ float to(T x) { return static_cast<float>(x); }
+  static __device__ __forceinline__ T from(float a) { return static_cast<T>(a); }
+};
+template <>
+struct AccOf<double> {
+  using type = double;
+  static __device__ __forceinline__ double to(double x) { return x; }
+  static __device__ __forceinline__ double from(double a) { return a; }
+};
+template <>
+struct AccOf<__half> {
+  using type = float;
+  static __device__ __forceinline__ float to(__half x) { return __half2float(x); }
+  static __device__ __forceinline__ __half from(float a) { return __float2half(a); }
+};
+template <>
+struct AccOf<__hip_bfloat16> {
+  using type = float;
+  static __device__ __forceinline__ float to(__hip_bfloat16 x) { return __bfloat162float(x); }
+  static __device__ __forceinline__ __hip_bfloat16 from(float a) { return __float2bfloat16(a); }
+};
+
+template <typename T, int VEC>
+struct alignas(sizeof(T) * VEC) Pack {
+  T v[VEC];
+};
+
+// elements per 16-byte access
+template <typename T>
+constexpr int vec_width() {
+  return 16 / sizeof(T);
+}
+
+inline int grid_for(long nitems) {
+  long blocks = (nitems + kThreads - 1) / kThreads;
+  // >> 256 workgroups to fill all 8 XCDs; cap so the grid-stride loop runs
+  if (blocks > 65535L * 8) blocks = 65535L * 8;
+  if (blocks < 1) blocks = 1;
+  return static_cast<int>(blocks);
+}
+
+// ---------------------------------------------------------------------------
+// out = self_w * self + sum_k w[k] * gathered[k*numel + i]
+// ---------------------------------------------------------------------------
+
+template <typename T, int VEC>
+__global__ __launch_bounds__(kThreads) void weighted_combine_k(
+    T* __restrict__ out, const T* __restrict__ self, float self_w,
+    const T* __restrict__ gathered, WeightsArg warg, int n_nbr, long numel) {
+  using Acc = typename AccOf<T>::type;
+  const long nvec = numel / VEC;
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> sv = *reinterpret_cast<const Pack<T, VEC>*>(self + base);
+    Acc acc[VEC];
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) acc[v] = static_cast<Acc>(self_w) * AccOf<T>::to(sv.v[v]);
+    for (int k = 0; k < n_nbr; ++k) {
+      Pack<T, VEC> gv =
+          *reinterpret_cast<const Pack<T, VEC>*>(gathered + static_cast<long>(k) * numel + base);
+      const Acc wk = static_cast<Acc>(warg.w[k]);
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) acc[v] += wk * AccOf<T>::to(gv.v[v]);
+    }
+    Pack<T, VEC> ov;
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) ov.v[v] = AccOf<T>::from(acc[v]);
+    *reinterpret_cast<Pack<T, VEC>*>(out + base) = ov;
+  }
+  // scalar tail
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    Acc acc = static_cast<Acc>(self_w) * AccOf<T>::to(self[t]);
+    for (int k = 0; k < n_nbr; ++k)
+      acc += static_cast<Acc>(warg.w[k]) * AccOf<T>::to(gathered[static_cast<long>(k) * numel + t]);
+    out[t] = AccOf<T>::from(acc);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dst = w * src   and   dst += w * src   (dst may be peer memory over xGMI)
+// ---------------------------------------------------------------------------
+
+template <typename T, int VEC, bool ACCUM>
+__global__ __launch_bounds__(kThreads) void scale_put_k(
+    T* __restrict__ dst, const T* __restrict__ src, float w, long numel) {
+  using Acc = typename AccOf<T>::type;
+  const long nvec = numel / VEC;
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> sv = *reinterpret_cast<const Pack<T, VEC>*>(src + base);
+    Pack<T, VEC> dv;
+    if (ACCUM) dv = *reinterpret_cast<const Pack<T, VEC>*>(dst + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      Acc x = static_cast<Acc>(w) * AccOf<T>::to(sv.v[v]);
+      if (ACCUM) x += AccOf<T>::to(dv.v[v]);
+      dv.v[v] = AccOf<T>::from(x);
+    }
+    *reinterpret_cast<Pack<T, VEC>*>(dst + base) = dv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    Acc x = static_cast<Acc>(w) * AccOf<T>::to(src[t]);
+    if (ACCUM) x += AccOf<T>::to(dst[t]);
+    dst[t] = AccOf<T>::from(x);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// in-place scaling (reference cuda_kernels.cu scale_buffer)
+// ---------------------------------------------------------------------------
+
+template <typename T, int VEC>
+__global__ __launch_bounds__(kThreads) void scale_inplace_k(T* __restrict__ buf,
+                                                            float f, long numel) {
+  using Acc = typename AccOf<T>::type;
+  const long nvec = numel / VEC;
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> dv = *reinterpret_cast<const Pack<T, VEC>*>(buf + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v)
+      dv.v[v] = AccOf<T>::from(static_cast<Acc>(f) * AccOf<T>::to(dv.v[v]));
+    *reinterpret_cast<Pack<T, VEC>*>(buf + base) = dv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride)
+    buf[t] = AccOf<T>::from(static_cast<Acc>(f) * AccOf<T>::to(buf[t]));
+}
+
+// ---------------------------------------------------------------------------
+// fused neighbor-average + SGD(momentum) step over one flat bucket:
+//   p   = self_w*p + sum_k w[k]*gathered[k]        (skip when n_nbr < 0)
+//   g   = grad + wd*p
+//   m   = mu*m + g            (when momentum buffer given)
+//   p  -= lr * (nesterov ? g + mu*m : (mu!=0 ? m : g))
+// One pass over HBM instead of average kernel + torch optimizer kernels.
+// ---------------------------------------------------------------------------
+
+template <typename T, int VEC, bool HAS_MOM>
+__global__ __launch_bounds__(kThreads) void combine_sgd_k(
+    T* __restrict__ p, float self_w, const T* __restrict__ gathered,
+    WeightsArg warg, int n_nbr, const T* __restrict__ grad, T* __restrict__ mom,
+    float lr, float mu, float wd, float dampening, int nesterov, long numel) {
+  using Acc = typename AccOf<T>::type;
+  const long nvec = numel / VEC;
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> pv = *reinterpret_cast<const Pack<T, VEC>*>(p + base);
+    Acc acc[VEC];
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) acc[v] = static_cast<Acc>(self_w) * AccOf<T>::to(pv.v[v]);
+    for (int k = 0; k < n_nbr; ++k) {
+      Pack<T, VEC> gv =
+          *reinterpret_cast<const Pack<T, VEC>*>(gathered + static_cast<long>(k) * numel + base);
+      const Acc wk = static_cast<Acc>(warg.w[k]);
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) acc[v] += wk * AccOf<T>::to(gv.v[v]);
+    }
+    Pack<T, VEC> grv = *reinterpret_cast<const Pack<T, VEC>*>(grad + base);
+    Pack<T, VEC> mv;
+    if (HAS_MOM) mv = *reinterpret_cast<const Pack<T, VEC>*>(mom + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      Acc g = AccOf<T>::to(grv.v[v]) + static_cast<Acc>(wd) * acc[v];
+      Acc upd = g;
+      if (HAS_MOM) {
+        Acc m = static_cast<Acc>(mu) * AccOf<T>::to(mv.v[v]) +
+                (static_cast<Acc>(1) - static_cast<Acc>(dampening)) * g;
+        mv.v[v] = AccOf<T>::from(m);
+        upd = nesterov ? g + static_cast<Acc>(mu) * m : m;
+      }
+      acc[v] -= static_cast<Acc>(lr) * upd;
+      pv.v[v] = AccOf<T>::from(acc[v]);
+    }
+    *reinterpret_cast<Pack<T, VEC>*>(p + base) = pv;
+    if (HAS_MOM) *reinterpret_cast<Pack<T, VEC>*>(mom + base) = mv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    Acc acc = static_cast<Acc>(self_w) * AccOf<T>::to(p[t]);
+    for (int k = 0; k < n_nbr; ++k)
+      acc += static_cast<Acc>(warg.w[k]) * AccOf<T>::to(gathered[static_cast<long>(k) * numel + t]);
+    Acc g = AccOf<T>::to(grad[t]) + static_cast<Acc>(wd) * acc;
+    Acc upd = g;
+    if (HAS_MOM) {
+      Acc m = static_cast<Acc>(mu) * AccOf<T>::to(mom[t]) +
+              (static_cast<Acc>(1) - static_cast<Acc>(dampening)) * g;
+      mom[t] = AccOf<T>::from(m);
+      upd = nesterov ? g + static_cast<Acc>(mu) * m : m;
+    }
+    p[t] = AccOf<T>::from(acc - static_cast<Acc>(lr) * upd);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fused neighbor-average + Adam step over one flat bucket (fp32 state)
+// ---------------------------------------------------------------------------
+
+template <typename T, int VEC>
+__global__ __launch_bounds__(kThreads) void combine_adam_k(
+    T* __restrict__ p, float self_w, const T* __restrict__ gathered,
+    WeightsArg warg, int n_nbr, const T* __restrict__ grad,
+    float* __restrict__ exp_avg, float* __restrict__ exp_avg_sq, float lr,
+    float beta1, float beta2, float eps, float wd, float bias1, float bias2,
+    long numel) {
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  long t = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; t < numel; t += stride) {
+    float acc = self_w * AccOf<T>::to(p[t]);
+    for (int k = 0; k < n_nbr; ++k)
+      acc += warg.w[k] * AccOf<T>::to(gathered[static_cast<long>(k) * numel + t]);
+    float g = AccOf<T>::to(grad[t]) + wd * acc;
+    float m = beta1 * exp_avg[t] + (1.f - beta1) * g;
+    float v = beta2 * exp_avg_sq[t] + (1.f - beta2) * g * g;
+    exp_avg[t] = m;
+    exp_avg_sq[t] = v;
+    const float denom = sqrtf(v / bias2) + eps;
+    p[t] = AccOf<T>::from(acc - lr * (m / bias1) / denom);
+  }
+}
+
+enum BfDtype : int { kF32 = 0, kF64 = 1, kF16 = 2, kBF16 = 3 };
+
+template <template <typename> class Fn, typename... Args>
+hipError_t dispatch_dtype(int dtype, Args&&... args) {
+  switch (dtype) {
+    case kF32:
+      return Fn<float>::run(std::forward<Args>(args)...);
+    case kF64:
+      return Fn<double>::run(std::forward<Args>(args)...);
+    case kF16:
+      return Fn<__half>::run(std::forward<Args>(args)...);
+    case kBF16:
+      return Fn<__hip_bfloat16>::run(std::forward<Args>(args)...);
+    default:
+      return hipErrorInvalidValue;
+  }
+}
+
+template <typename T>
+bool vec_ok(const void* p, long numel) {
+  // 16-byte vector path needs 16B-aligned base pointers; slice strides of
+  // the gathered block are numel*sizeof(T) so numel must keep alignment too
+  return (reinterpret_cast<uintptr_t>(p) % 16 == 0) &&
+         ((numel * sizeof(T)) % 16 == 0);
+}
+
+template <typename T>
+struct CombineLauncher {
+  static hipError_t run(void* out, const void* self, float self_w,
+                        const void* gathered, const float* w, int n_nbr,
+                        long numel, hipStream_t stream) {
+    constexpr int V = vec_width<T>();
+    // chunk neighbors by kMaxNbrPerLaunch; later chunks accumulate onto out
+    int done = 0;
+    float cur_self_w = self_w;
+    const T* cur_self = static_cast<const T*>(self);
+    do {
+      WeightsArg warg{};
+      const int n = (n_nbr - done) > kMaxNbrPerLaunch ? kMaxNbrPerLaunch : (n_nbr - done);
+      for (int k = 0; k < n; ++k) warg.w[k] = w[done + k];
+      const T* g = static_cast<const T*>(gathered) + static_cast<long>(done) * numel;
+      const bool vec = vec_ok<T>(out, numel) && vec_ok<T>(self, numel) &&
+                       (n == 0 || vec_ok<T>(g, numel));
+      const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
+      if (vec) {
+        hipLaunchKernelGGL((weighted_combine_k<T, V>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(out), cur_self, cur_self_w, g,
+                           warg, n, numel);
+      } else {
+        hipLaunchKernelGGL((weighted_combine_k<T, 1>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(out), cur_self, cur_self_w, g,
+                           warg, n, numel);
+      }
+      BF_CHECK_HIP(hipGetLastError());
+      done += n;
+      cur_self = static_cast<T*>(out);
+      cur_self_w = 1.0f;
+    } while (done < n_nbr);
+    return hipSuccess;
+  }
+};
+
+template <typename T>
+struct ScalePutLauncher {
+  static hipError_t run(void* dst, const void* src, float w, long numel,
+                        bool accum, hipStream_t stream) {
+    constexpr int V = vec_width<T>();
+    const bool vec = vec_ok<T>(dst, numel) && vec_ok<T>(src, numel);
+    const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
+    if (accum) {
+      if (vec)
+        hipLaunchKernelGGL((scale_put_k<T, V, true>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(dst), static_cast<const T*>(src), w, numel);
+      else
+        hipLaunchKernelGGL((scale_put_k<T, 1, true>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(dst), static_cast<const T*>(src), w, numel);
+    } else {
+      if (vec)
+        hipLaunchKernelGGL((scale_put_k<T, V, false>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(dst), static_cast<const T*>(src), w, numel);
+      else
+        hipLaunchKernelGGL((scale_put_k<T, 1, false>), dim3(grid), dim3(kThreads), 0,
+                           stream, static_cast<T*>(dst), static_cast<const T*>(src), w, numel);
+    }
+    return hipGetLastError();
+  }
+};
+
+template <typename T>
+struct ScaleInplaceLauncher {
+  static hipError_t run(void* buf, float f, long numel, hipStream_t stream) {
+    constexpr int V = vec_width<T>();
+    const bool vec = vec_ok<T>(buf, numel);
+    const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
+    if (vec)
+      hipLaunchKernelGGL((scale_inplace_k<T, V>), dim3(grid), dim3(kThreads), 0,
+                         stream, static_cast<T*>(buf), f, numel);
+    else
+      hipLaunchKernelGGL((scale_inplace_k<T, 1>), dim3(grid), dim3(kThreads), 0,
+                         stream, static_cast<T*>(buf), f, numel);
+    return hipGetLastError();
+  }
+};
+
+template <typename T>
+struct CombineSgdLauncher {
+  static hipError_t run(void* p, float self_w, const void* gathered,
+                        const float* w, int n_nbr, const void* grad, void* mom,
+                        float lr, float mu, float wd, float dampening,
+                        int nesterov, long numel, hipStream_t stream) {
+    constexpr int V = vec_width<T>();
+    if (n_nbr > kMaxNbrPerLaunch) {
+      // rare: pre-combine the overflow neighbors, then fused step on the rest
+      const int overflow = n_nbr - kMaxNbrPerLaunch;
+      BF_CHECK_HIP((CombineLauncher<T>::run(p, p, self_w, gathered, w, overflow,
+                                            numel, stream)));
+      gathered = static_cast<const T*>(gathered) + static_cast<long>(overflow) * numel;
+      w += overflow;
+      n_nbr = kMaxNbrPerLaunch;
+      self_w = 1.0f;
+    }
+    WeightsArg warg{};
+    for (int k = 0; k < n_nbr; ++k) warg.w[k] = w[k];
+    const bool vec = vec_ok<T>(p, numel) && vec_ok<T>(grad, numel) &&
+                     (n_nbr == 0 || vec_ok<T>(gathered, numel)) &&
+                     (mom == nullptr || vec_ok<T>(mom, numel));
+    const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
+    const bool has_mom = mom != nullptr;
+#define BF_LAUNCH_SGD(VV, MM)                                                  \
+  hipLaunchKernelGGL((combine_sgd_k<T, VV, MM>), dim3(grid), dim3(kThreads), 0,\
+                     stream, static_cast<T*>(p), self_w,                       \
+                     static_cast<const T*>(gathered), warg, n_nbr,             \
+                     static_cast<const T*>(grad), static_cast<T*>(mom), lr,    \
+                     mu, wd, dampening, nesterov, numel)
+    if (vec) {
+      if (has_mom)
+        BF_LAUNCH_SGD(V, true);
+      else
+        BF_LAUNCH_SGD(V, false);
+    } else {
+      if (has_mom)
+        BF_LAUNCH_SGD(1, true);
+      else
+        BF_LAUNCH_SGD(1, false);
+    }
+#undef BF_LAUNCH_SGD
+    return hipGetLastError();
+  }
+};
+
+template <typename T>
+struct CombineAdamLauncher {
+  static hipError_t run(void* p, float self_w, const void* gathered,
+                        const float* w, int n_nbr, const void* grad,
+                        float* exp_avg, float* exp_avg_sq, float lr, float beta1,
+                        float beta2, float eps, float wd, float bias1,
+                        float bias2, long numel, hipStream_t stream) {
+    if (n_nbr > kMaxNbrPerLaunch) {
+      const int overflow = n_nbr - kMaxNbrPerLaunch;
+      BF_CHECK_HIP((CombineLauncher<T>::run(p, p, self_w, gathered, w, overflow,
+                                            numel, stream)));
+      gathered = static_cast<const T*>(gathered) + static_cast<long>(overflow) * numel;
+      w += overflow;
+      n_nbr = kMaxNbrPerLaunch;
+      self_w = 1.0f;
+    }
+    WeightsArg warg{};
+    for (int k = 0; k < n_nbr; ++k) warg.w[k] = w[k];
+    const int grid = grid_for(numel);
+    hipLaunchKernelGGL((combine_adam_k<T, 1>), dim3(grid), dim3(kThreads), 0,
+                       stream, static_cast<T*>(p), self_w,
+                       static_cast<const T*>(gathered), warg, n_nbr,
+                       static_cast<const T*>(grad), exp_avg, exp_avg_sq, lr,
+                       beta1, beta2, eps, wd, bias1, bias2, numel);
+    return hipGetLastError();
+  }
+};
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// C interface (called from bindings.cpp)
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+hipError_t bf_weighted_combine(void* out, const void* self, float self_w,
+                               const void* gathered, const float* w, int n_nbr,
+                               long numel, int dtype, hipStream_t stream) {
+  return dispatch_dtype<CombineLauncher>(dtype, out, self, self_w, gathered, w,
+                                         n_nbr, numel, stream);
+}
+
+hipError_t bf_scale_put(void* dst, const void* src, float w, long numel,
+                        int dtype, bool accum, hipStream_t stream) {
+  return dispatch_dtype<ScalePutLauncher>(dtype, dst, src, w, numel, accum,
+                                          stream);
+}
+
+hipError_t bf_scale_inplace(void* buf, float f, long numel, int dtype,
+                            hipStream_t stream) {
+  return dispatch_dtype<ScaleInplaceLauncher>(dtype, buf, f, numel, stream);
+}
+
+hipError_t bf_combine_sgd(void* p, float self_w, const void* gathered,
+                          const float* w, int n_nbr, const void* grad,
+                          void* mom, float lr, float mu, float wd,
+                          float dampening, int nesterov, long numel, int dtype,
+                          hipStream_t stream) {
+  return dispatch_dtype<CombineSgdLauncher>(dtype, p, self_w, gathered, w,
+                                            n_nbr, grad, mom, lr, mu, wd,
+                                            dampening, nesterov, numel, stream);
+}
+
+hipError_t bf_combine_adam(void* p, float self_w, const void* gathered,
+                           const float* w, int n_nbr, const void* grad,
+                           float* exp_avg, float* exp_avg_sq, float lr,
+                           float beta1, float beta2, float eps, float wd,
+                           float bias1, float bias2, long numel, int dtype,
+                           hipStream_t stream) {
+  return dispatch_dtype<CombineAdamLauncher>(dtype, p, self_w, gathered, w,
+                                             n_nbr, grad, exp_avg, exp_avg_sq,
+                                             lr, beta1, beta2, eps, wd, bias1,
+                                             bias2, numel, stream);
+}
+
+}  // extern "C"

@@ -100,9 +100,13 @@ class BlueFogContext:
             self._device = torch.device("cpu")
 
         if not dist.is_initialized():
-            backend = "cpu:gloo,cuda:nccl" if use_cuda else "gloo"
+            # BLUEFOG_BACKEND overrides (e.g. "gloo" to run several ranks on
+            # one GPU in tests — RCCL forbids two ranks on one device)
+            backend = os.environ.get(
+                "BLUEFOG_BACKEND", "cpu:gloo,cuda:nccl" if use_cuda else "gloo"
+            )
             kwargs = {}
-            if use_cuda:
+            if use_cuda and "nccl" in backend:
                 kwargs["device_id"] = self._device
             dist.init_process_group(
                 backend=backend,

@@ -109,6 +109,22 @@ def _check_named_parameters(optimizer, model):
 # ---------------------------------------------------------------------------
 
 
+def _is_dense(t: torch.Tensor) -> bool:
+    """True iff the tensor covers its storage span exactly once under some
+    permutation of dims (contiguous, channels_last, ...)."""
+    if t.numel() == 0:
+        return True
+    dims = sorted(range(t.dim()), key=lambda d: t.stride(d))
+    expect = 1
+    for d in dims:
+        if t.shape[d] == 1:
+            continue
+        if t.stride(d) != expect:
+            return False
+        expect *= t.shape[d]
+    return expect == t.numel()
+
+
 class _FlatBuckets:
     """Flatten parameters into persistent contiguous per-(device,dtype)
     buckets and rebind each ``p.data`` to a view of its bucket."""
@@ -147,8 +163,15 @@ class _FlatBuckets:
             for _, p in named_plist:
                 n = p.numel()
                 seg = flat.narrow(0, off, n)
-                seg.copy_(p.data.reshape(-1))
-                p.data = seg.view(p.shape)
+                if _is_dense(p.data):
+                    # preserve the physical layout (e.g. channels_last conv
+                    # weights) inside the bucket via a strided view
+                    view = torch.as_strided(seg, p.shape, p.data.stride())
+                    view.copy_(p.data)
+                    p.data = view
+                else:
+                    seg.copy_(p.data.reshape(-1))
+                    p.data = seg.view(p.shape)
                 off += n
         self.buckets.append(
             {
