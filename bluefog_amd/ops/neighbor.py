@@ -72,6 +72,38 @@ def _post_neighbor_exchange(
     return works, gathered, keep_alive
 
 
+def post_neighbor_exchange_raw(
+    tensor: torch.Tensor,
+    self_weight: Optional[float],
+    src_weights,
+    dst_weights,
+    enable_topo_check: bool,
+):
+    """Low-level entry for fused consumers (the AWC optimizer's fused
+    average+step path): resolve weights, post the batched RCCL exchange and
+    return (works, gathered, src_weight_list, self_weight, keep_alive)
+    WITHOUT scheduling any post-processing — the caller fuses its own."""
+    tensor = tensor.detach()
+    if not tensor.is_contiguous():
+        tensor = tensor.contiguous()
+    (
+        self_weight,
+        src_weights,
+        dst_weights,
+        dynamic_enabled,
+        dst_weighting_enabled,
+    ) = engine.resolve_recv_weights(self_weight, src_weights, dst_weights)
+    src_ranks = list(src_weights.keys())
+    dst_ranks = list(dst_weights.keys())
+    if dynamic_enabled and enable_topo_check:
+        engine.check_src_dst_consistency(src_ranks, dst_ranks, "neighbor_allreduce")
+    works, gathered, keep_alive = _post_neighbor_exchange(
+        tensor, src_ranks, dst_ranks, dst_weights, dst_weighting_enabled
+    )
+    weights = [src_weights[r] for r in src_ranks]
+    return works, gathered, weights, self_weight, keep_alive
+
+
 def _neighbor_allreduce_nonblocking_impl(
     tensor: torch.Tensor,
     self_weight: Optional[float],

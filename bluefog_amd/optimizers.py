@@ -316,6 +316,7 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
         self._use_timeline = False
 
         self._buckets = _FlatBuckets(named_parameters, fusion_threshold_bytes())
+        self._fused = self._init_fused_mode()
         if os.getenv("BLUEFOG_TIMELINE"):
             self.turn_on_timeline()
         if _bf().size() > 1:
@@ -323,6 +324,110 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
 
     def _bluefog_base_step(self, closure=None):
         return super(self.__class__, self).step(closure)
+
+    # ------------------------------------------------------------------
+    # fused average+step: one CDNA4 kernel per bucket does
+    #   p = self_w*p + sum_k w_k*recv_k  followed by the SGD/Adam update
+    # in a single pass over HBM (the reference runs torch slice math plus
+    # the base optimizer's kernels instead). Gradients are flattened into
+    # per-bucket buffers so the kernel sees one contiguous grad stream.
+    # ------------------------------------------------------------------
+    def _init_fused_mode(self):
+        if os.environ.get("BLUEFOG_FUSED_STEP", "1") in ("0", "false"):
+            return None
+        if not self._buckets.buckets:
+            return None
+        dev = self._buckets.buckets[0]["flat"].device
+        if dev.type != "cuda":
+            return None
+        from bluefog_amd.ops import hip_ext
+
+        if not hip_ext.has_extension():
+            return None
+        if len(self.param_groups) != 1:
+            return None
+        base = type(self).__mro__[1]
+        group = self.param_groups[0]
+        if base is torch.optim.SGD and group.get("dampening", 0) == 0:
+            kind = "sgd"
+        elif base is torch.optim.Adam and not group.get("amsgrad", False):
+            kind = "adam"
+        else:
+            return None
+        if self._communication_type not in (
+            CommunicationType.neighbor_allreduce,
+            CommunicationType.empty,
+        ):
+            return None
+        # flatten gradients: p.grad becomes a view of the bucket's grad
+        # buffer with p's physical layout, so autograd accumulates in place
+        with torch.no_grad():
+            for b in self._buckets.buckets:
+                b["grad_flat"] = torch.zeros_like(b["flat"])
+                off = 0
+                for p in b["params"]:
+                    n = p.numel()
+                    seg = b["grad_flat"].narrow(0, off, n)
+                    p.grad = torch.as_strided(seg, p.shape, p.data.stride())
+                    off += n
+                if kind == "sgd":
+                    b["momentum_flat"] = None  # lazily allocated
+                else:
+                    b["exp_avg"] = torch.zeros(
+                        b["flat"].numel(), dtype=torch.float32, device=dev
+                    )
+                    b["exp_avg_sq"] = torch.zeros(
+                        b["flat"].numel(), dtype=torch.float32, device=dev
+                    )
+                    b["adam_step"] = 0
+        logger.debug("bluefog_amd: fused %s average+step enabled", kind)
+        return kind
+
+    def _fused_apply(self, b, works, gathered, weights, self_weight):
+        """Make the current stream wait for the bucket's communication, then
+        run the fused average+optimizer-step kernel on it."""
+        from bluefog_amd import _C
+
+        for w in works:
+            w.wait()  # stream-level wait on RCCL work
+        group = self.param_groups[0]
+        flat, grad = b["flat"], b["grad_flat"]
+        gathered = gathered if gathered is not None else flat
+        if self._fused == "sgd":
+            momentum = group["momentum"]
+            if momentum != 0 and b["momentum_flat"] is None:
+                b["momentum_flat"] = torch.zeros_like(flat)
+            _C.weighted_combine_sgd(
+                flat,
+                float(self_weight),
+                gathered,
+                [float(w) for w in weights],
+                grad,
+                b["momentum_flat"] if momentum != 0 else flat.new_empty(0),
+                float(group["lr"]),
+                float(momentum),
+                float(group["weight_decay"]),
+                float(group.get("dampening", 0.0)),
+                bool(group.get("nesterov", False)),
+            )
+        else:
+            b["adam_step"] += 1
+            beta1, beta2 = group["betas"]
+            _C.weighted_combine_adam(
+                flat,
+                float(self_weight),
+                gathered,
+                [float(w) for w in weights],
+                grad,
+                b["exp_avg"],
+                b["exp_avg_sq"],
+                float(group["lr"]),
+                float(beta1),
+                float(beta2),
+                float(group["eps"]),
+                float(group["weight_decay"]),
+                int(b["adam_step"]),
+            )
 
     def _register_hooks(self):
         for model in self._models:
@@ -346,6 +451,24 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
 
     def _launch_communication(self):
         bf = _bf()
+        if self._fused and self._communication_type == CommunicationType.neighbor_allreduce:
+            from bluefog_amd.ops import neighbor
+
+            for b in self._buckets.buckets:
+                works, gathered, weights, self_w, keep = (
+                    neighbor.post_neighbor_exchange_raw(
+                        b["flat"],
+                        self.self_weight,
+                        self.src_weights,
+                        self.dst_weights,
+                        self.enable_topo_check,
+                    )
+                )
+                self._handles[b["name"]] = (
+                    b,
+                    ("fused", works, gathered, weights, self_w, keep),
+                )
+            return
         for b in self._buckets.buckets:
             name = b["name"]
             if self._communication_type == CommunicationType.allreduce:
@@ -393,7 +516,11 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
         bf = _bf()
         with torch.no_grad():
             for name, (b, handle) in self._handles.items():
-                if handle is not None:
+                if isinstance(handle, tuple) and handle and handle[0] == "fused":
+                    _, works, gathered, weights, self_w, keep = handle
+                    self._fused_apply(b, works, gathered, weights, self_w)
+                    self._fused_step_done = True
+                elif handle is not None:
                     output = bf.synchronize(handle)
                     b["flat"].copy_(output)
             self._reduce_delay = self._num_steps_per_communication
@@ -418,7 +545,29 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
                 )
             self.synchronize()
         self._synchronized = False
+        if self._fused is not None:
+            # the fused kernel applied avg+update for comm iterations (flag
+            # set in synchronize); on communication-free iterations run the
+            # same kernel as a plain optimizer step (no neighbors)
+            if not getattr(self, "_fused_step_done", False):
+                with torch.no_grad():
+                    for b in self._buckets.buckets:
+                        self._fused_apply(b, [], None, [], 1.0)
+            self._fused_step_done = False
+            if closure is not None:
+                with torch.enable_grad():
+                    return closure()
+            return None
         return super(self.__class__, self).step(closure)
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self._fused is not None:
+            # gradients are views into per-bucket flat buffers; zero those
+            # (set_to_none would detach autograd from the flat storage)
+            for b in self._buckets.buckets:
+                b["grad_flat"].zero_()
+            return None
+        return super(self.__class__, self).zero_grad(set_to_none)
 
 
 # ---------------------------------------------------------------------------
