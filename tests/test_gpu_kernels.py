@@ -41,15 +41,17 @@ def test_weighted_combine_numerics(dev, dtype, atol, numel):
     self_w = 0.25
     out = torch.empty_like(self_t)
     he.weighted_combine(out, self_t, self_w, gathered, weights)
-    # plain fp32 torch reference
-    acc = self_w * self_t.float()
+    # plain torch reference in the kernel's accumulation precision
+    # (fp32 for f32/f16/bf16 inputs, fp64 for f64)
+    acc_dtype = torch.float64 if dtype == torch.float64 else torch.float32
+    acc = self_w * self_t.to(acc_dtype)
     for k in range(n_nbr):
-        acc += weights[k] * gathered[k * numel : (k + 1) * numel].float()
+        acc += weights[k] * gathered[k * numel : (k + 1) * numel].to(acc_dtype)
     torch.cuda.synchronize()
-    assert torch.allclose(out.float(), acc.to(dtype).float(), atol=atol), (
+    assert torch.allclose(out.to(acc_dtype), acc.to(dtype).to(acc_dtype), atol=atol), (
         dtype,
         numel,
-        (out.float() - acc).abs().max().item(),
+        (out.to(acc_dtype) - acc).abs().max().item(),
     )
 
 
