@@ -15,23 +15,23 @@
 #include <vector>
 
 extern "C" {
-hipError_t bf_weighted_combine(void* out, const void* self, float self_w,
-                               const void* gathered, const float* w, int n_nbr,
+hipError_t bf_weighted_combine(void* out, const void* self, double self_w,
+                               const void* gathered, const double* w, int n_nbr,
                                long numel, int dtype, hipStream_t stream);
-hipError_t bf_scale_put(void* dst, const void* src, float w, long numel,
+hipError_t bf_scale_put(void* dst, const void* src, double w, long numel,
                         int dtype, bool accum, hipStream_t stream);
-hipError_t bf_scale_inplace(void* buf, float f, long numel, int dtype,
+hipError_t bf_scale_inplace(void* buf, double f, long numel, int dtype,
                             hipStream_t stream);
-hipError_t bf_combine_sgd(void* p, float self_w, const void* gathered,
-                          const float* w, int n_nbr, const void* grad,
-                          void* mom, float lr, float mu, float wd,
-                          float dampening, int nesterov, long numel, int dtype,
+hipError_t bf_combine_sgd(void* p, double self_w, const void* gathered,
+                          const double* w, int n_nbr, const void* grad,
+                          void* mom, double lr, double mu, double wd,
+                          double dampening, int nesterov, long numel, int dtype,
                           hipStream_t stream);
-hipError_t bf_combine_adam(void* p, float self_w, const void* gathered,
-                           const float* w, int n_nbr, const void* grad,
-                           float* exp_avg, float* exp_avg_sq, float lr,
-                           float beta1, float beta2, float eps, float wd,
-                           float bias1, float bias2, long numel, int dtype,
+hipError_t bf_combine_adam(void* p, double self_w, const void* gathered,
+                           const double* w, int n_nbr, const void* grad,
+                           float* exp_avg, float* exp_avg_sq, double lr,
+                           double beta1, double beta2, double eps, double wd,
+                           double bias1, double bias2, long numel, int dtype,
                            hipStream_t stream);
 }
 
@@ -73,12 +73,11 @@ void weighted_combine(at::Tensor output, at::Tensor self, double self_weight,
                 "gathered must hold one self-shaped slice per weight");
     TORCH_CHECK(gathered.scalar_type() == self.scalar_type(), "dtype mismatch");
   }
-  std::vector<float> w(weights.begin(), weights.end());
   check_hip(bf_weighted_combine(output.data_ptr(), self.data_ptr(),
-                                static_cast<float>(self_weight),
+                                self_weight,
                                 n ? gathered.data_ptr() : self.data_ptr(),
-                                w.data(), n, self.numel(), dtype_code(self),
-                                current_stream()),
+                                weights.data(), n, self.numel(),
+                                dtype_code(self), current_stream()),
             "weighted_combine");
 }
 
@@ -86,8 +85,7 @@ void scale_put(at::Tensor dst, at::Tensor src, double weight) {
   TORCH_CHECK(dst.is_contiguous() && src.is_contiguous(), "contiguous only");
   TORCH_CHECK(dst.numel() == src.numel(), "numel mismatch");
   TORCH_CHECK(dst.scalar_type() == src.scalar_type(), "dtype mismatch");
-  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(),
-                         static_cast<float>(weight), src.numel(),
+  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(), weight, src.numel(),
                          dtype_code(src), /*accum=*/false, current_stream()),
             "scale_put");
 }
@@ -96,16 +94,15 @@ void accum_put(at::Tensor dst, at::Tensor src, double weight) {
   TORCH_CHECK(dst.is_contiguous() && src.is_contiguous(), "contiguous only");
   TORCH_CHECK(dst.numel() == src.numel(), "numel mismatch");
   TORCH_CHECK(dst.scalar_type() == src.scalar_type(), "dtype mismatch");
-  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(),
-                         static_cast<float>(weight), src.numel(),
+  check_hip(bf_scale_put(dst.data_ptr(), src.data_ptr(), weight, src.numel(),
                          dtype_code(src), /*accum=*/true, current_stream()),
             "accum_put");
 }
 
 void scale_inplace(at::Tensor buf, double factor) {
   TORCH_CHECK(buf.is_contiguous(), "contiguous only");
-  check_hip(bf_scale_inplace(buf.data_ptr(), static_cast<float>(factor),
-                             buf.numel(), dtype_code(buf), current_stream()),
+  check_hip(bf_scale_inplace(buf.data_ptr(), factor, buf.numel(),
+                             dtype_code(buf), current_stream()),
             "scale_inplace");
 }
 
@@ -117,20 +114,17 @@ void weighted_combine_sgd(at::Tensor param, double self_weight,
   TORCH_CHECK(param.is_contiguous() && grad.is_contiguous(), "contiguous only");
   TORCH_CHECK(param.numel() == grad.numel(), "param/grad numel mismatch");
   const int n = static_cast<int>(weights.size());
-  std::vector<float> w(weights.begin(), weights.end());
   void* mom = nullptr;
   if (momentum_buf.defined() && momentum_buf.numel() > 0) {
     TORCH_CHECK(momentum_buf.numel() == param.numel(), "momentum numel");
     mom = momentum_buf.data_ptr();
   }
   check_hip(
-      bf_combine_sgd(param.data_ptr(), static_cast<float>(self_weight),
-                     n ? gathered.data_ptr() : param.data_ptr(), w.data(), n,
-                     grad.data_ptr(), mom, static_cast<float>(lr),
-                     static_cast<float>(momentum),
-                     static_cast<float>(weight_decay),
-                     static_cast<float>(dampening), nesterov ? 1 : 0,
-                     param.numel(), dtype_code(param), current_stream()),
+      bf_combine_sgd(param.data_ptr(), self_weight,
+                     n ? gathered.data_ptr() : param.data_ptr(),
+                     weights.data(), n, grad.data_ptr(), mom, lr, momentum,
+                     weight_decay, dampening, nesterov ? 1 : 0, param.numel(),
+                     dtype_code(param), current_stream()),
       "weighted_combine_sgd");
 }
 
@@ -145,18 +139,15 @@ void weighted_combine_adam(at::Tensor param, double self_weight,
                   exp_avg_sq.scalar_type() == at::kFloat,
               "adam state must be fp32");
   const int n = static_cast<int>(weights.size());
-  std::vector<float> w(weights.begin(), weights.end());
-  const float bias1 = 1.0f - std::pow(static_cast<float>(beta1), step);
-  const float bias2 = 1.0f - std::pow(static_cast<float>(beta2), step);
+  const double bias1 = 1.0 - std::pow(beta1, static_cast<double>(step));
+  const double bias2 = 1.0 - std::pow(beta2, static_cast<double>(step));
   check_hip(
-      bf_combine_adam(param.data_ptr(), static_cast<float>(self_weight),
-                      n ? gathered.data_ptr() : param.data_ptr(), w.data(), n,
-                      grad.data_ptr(), exp_avg.data_ptr<float>(),
-                      exp_avg_sq.data_ptr<float>(), static_cast<float>(lr),
-                      static_cast<float>(beta1), static_cast<float>(beta2),
-                      static_cast<float>(eps), static_cast<float>(weight_decay),
-                      bias1, bias2, param.numel(), dtype_code(param),
-                      current_stream()),
+      bf_combine_adam(param.data_ptr(), self_weight,
+                      n ? gathered.data_ptr() : param.data_ptr(),
+                      weights.data(), n, grad.data_ptr(),
+                      exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(),
+                      lr, beta1, beta2, eps, weight_decay, bias1, bias2,
+                      param.numel(), dtype_code(param), current_stream()),
       "weighted_combine_adam");
 }
 

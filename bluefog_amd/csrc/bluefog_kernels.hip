@@ -43,8 +43,12 @@ namespace {
 constexpr int kThreads = 256;
 constexpr int kMaxNbrPerLaunch = 16;  // weights ride the kernarg segment
 
-struct WeightsArg {
-  float w[kMaxNbrPerLaunch];
+// Per-launch neighbor weights ride the kernarg segment in the kernel's
+// accumulation type (double for f64 tensors, float otherwise) so fp64
+// averaging is exact to the ulp of the inputs.
+template <typename A>
+struct WeightsArgT {
+  A w[kMaxNbrPerLaunch];
 };
 
 // Conversion traits: torch builds with __HIP_NO_HALF_CONVERSIONS__, so
@@ -99,8 +103,9 @@ inline int grid_for(long nitems) {
 
 template <typename T, int VEC>
 __global__ __launch_bounds__(kThreads) void weighted_combine_k(
-    T* __restrict__ out, const T* __restrict__ self, float self_w,
-    const T* __restrict__ gathered, WeightsArg warg, int n_nbr, long numel) {
+    T* __restrict__ out, const T* __restrict__ self,
+    typename AccOf<T>::type self_w, const T* __restrict__ gathered,
+    WeightsArgT<typename AccOf<T>::type> warg, int n_nbr, long numel) {
   using Acc = typename AccOf<T>::type;
   const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
@@ -139,7 +144,8 @@ __global__ __launch_bounds__(kThreads) void weighted_combine_k(
 
 template <typename T, int VEC, bool ACCUM>
 __global__ __launch_bounds__(kThreads) void scale_put_k(
-    T* __restrict__ dst, const T* __restrict__ src, float w, long numel) {
+    T* __restrict__ dst, const T* __restrict__ src,
+    typename AccOf<T>::type w, long numel) {
   using Acc = typename AccOf<T>::type;
   const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
@@ -170,8 +176,8 @@ __global__ __launch_bounds__(kThreads) void scale_put_k(
 // ---------------------------------------------------------------------------
 
 template <typename T, int VEC>
-__global__ __launch_bounds__(kThreads) void scale_inplace_k(T* __restrict__ buf,
-                                                            float f, long numel) {
+__global__ __launch_bounds__(kThreads) void scale_inplace_k(
+    T* __restrict__ buf, typename AccOf<T>::type f, long numel) {
   using Acc = typename AccOf<T>::type;
   const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
@@ -200,9 +206,13 @@ __global__ __launch_bounds__(kThreads) void scale_inplace_k(T* __restrict__ buf,
 
 template <typename T, int VEC, bool HAS_MOM>
 __global__ __launch_bounds__(kThreads) void combine_sgd_k(
-    T* __restrict__ p, float self_w, const T* __restrict__ gathered,
-    WeightsArg warg, int n_nbr, const T* __restrict__ grad, T* __restrict__ mom,
-    float lr, float mu, float wd, float dampening, int nesterov, long numel) {
+    T* __restrict__ p, typename AccOf<T>::type self_w,
+    const T* __restrict__ gathered,
+    WeightsArgT<typename AccOf<T>::type> warg, int n_nbr,
+    const T* __restrict__ grad, T* __restrict__ mom,
+    typename AccOf<T>::type lr, typename AccOf<T>::type mu,
+    typename AccOf<T>::type wd, typename AccOf<T>::type dampening,
+    int nesterov, long numel) {
   using Acc = typename AccOf<T>::type;
   const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
@@ -262,23 +272,28 @@ __global__ __launch_bounds__(kThreads) void combine_sgd_k(
 
 template <typename T, int VEC>
 __global__ __launch_bounds__(kThreads) void combine_adam_k(
-    T* __restrict__ p, float self_w, const T* __restrict__ gathered,
-    WeightsArg warg, int n_nbr, const T* __restrict__ grad,
-    float* __restrict__ exp_avg, float* __restrict__ exp_avg_sq, float lr,
-    float beta1, float beta2, float eps, float wd, float bias1, float bias2,
+    T* __restrict__ p, typename AccOf<T>::type self_w,
+    const T* __restrict__ gathered,
+    WeightsArgT<typename AccOf<T>::type> warg, int n_nbr,
+    const T* __restrict__ grad, float* __restrict__ exp_avg,
+    float* __restrict__ exp_avg_sq, typename AccOf<T>::type lr,
+    typename AccOf<T>::type beta1, typename AccOf<T>::type beta2,
+    typename AccOf<T>::type eps, typename AccOf<T>::type wd,
+    typename AccOf<T>::type bias1, typename AccOf<T>::type bias2,
     long numel) {
+  using Acc = typename AccOf<T>::type;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   long t = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
   for (; t < numel; t += stride) {
-    float acc = self_w * AccOf<T>::to(p[t]);
+    Acc acc = self_w * AccOf<T>::to(p[t]);
     for (int k = 0; k < n_nbr; ++k)
       acc += warg.w[k] * AccOf<T>::to(gathered[static_cast<long>(k) * numel + t]);
-    float g = AccOf<T>::to(grad[t]) + wd * acc;
-    float m = beta1 * exp_avg[t] + (1.f - beta1) * g;
-    float v = beta2 * exp_avg_sq[t] + (1.f - beta2) * g * g;
-    exp_avg[t] = m;
-    exp_avg_sq[t] = v;
-    const float denom = sqrtf(v / bias2) + eps;
+    Acc g = AccOf<T>::to(grad[t]) + wd * acc;
+    Acc m = beta1 * static_cast<Acc>(exp_avg[t]) + (static_cast<Acc>(1) - beta1) * g;
+    Acc v = beta2 * static_cast<Acc>(exp_avg_sq[t]) + (static_cast<Acc>(1) - beta2) * g * g;
+    exp_avg[t] = static_cast<float>(m);
+    exp_avg_sq[t] = static_cast<float>(v);
+    const Acc denom = sqrt(v / bias2) + eps;
     p[t] = AccOf<T>::from(acc - lr * (m / bias1) / denom);
   }
 }
@@ -311,18 +326,19 @@ bool vec_ok(const void* p, long numel) {
 
 template <typename T>
 struct CombineLauncher {
-  static hipError_t run(void* out, const void* self, float self_w,
-                        const void* gathered, const float* w, int n_nbr,
+  using Acc = typename AccOf<T>::type;
+  static hipError_t run(void* out, const void* self, double self_w,
+                        const void* gathered, const double* w, int n_nbr,
                         long numel, hipStream_t stream) {
     constexpr int V = vec_width<T>();
     // chunk neighbors by kMaxNbrPerLaunch; later chunks accumulate onto out
     int done = 0;
-    float cur_self_w = self_w;
+    Acc cur_self_w = static_cast<Acc>(self_w);
     const T* cur_self = static_cast<const T*>(self);
     do {
-      WeightsArg warg{};
+      WeightsArgT<Acc> warg{};
       const int n = (n_nbr - done) > kMaxNbrPerLaunch ? kMaxNbrPerLaunch : (n_nbr - done);
-      for (int k = 0; k < n; ++k) warg.w[k] = w[done + k];
+      for (int k = 0; k < n; ++k) warg.w[k] = static_cast<Acc>(w[done + k]);
       const T* g = static_cast<const T*>(gathered) + static_cast<long>(done) * numel;
       const bool vec = vec_ok<T>(out, numel) && vec_ok<T>(self, numel) &&
                        (n == 0 || vec_ok<T>(g, numel));
@@ -339,7 +355,7 @@ struct CombineLauncher {
       BF_CHECK_HIP(hipGetLastError());
       done += n;
       cur_self = static_cast<T*>(out);
-      cur_self_w = 1.0f;
+      cur_self_w = static_cast<Acc>(1);
     } while (done < n_nbr);
     return hipSuccess;
   }
@@ -347,8 +363,10 @@ struct CombineLauncher {
 
 template <typename T>
 struct ScalePutLauncher {
-  static hipError_t run(void* dst, const void* src, float w, long numel,
+  using Acc = typename AccOf<T>::type;
+  static hipError_t run(void* dst, const void* src, double w_in, long numel,
                         bool accum, hipStream_t stream) {
+    const Acc w = static_cast<Acc>(w_in);
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(dst, numel) && vec_ok<T>(src, numel);
     const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
@@ -373,7 +391,9 @@ struct ScalePutLauncher {
 
 template <typename T>
 struct ScaleInplaceLauncher {
-  static hipError_t run(void* buf, float f, long numel, hipStream_t stream) {
+  using Acc = typename AccOf<T>::type;
+  static hipError_t run(void* buf, double f_in, long numel, hipStream_t stream) {
+    const Acc f = static_cast<Acc>(f_in);
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(buf, numel);
     const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
@@ -389,9 +409,10 @@ struct ScaleInplaceLauncher {
 
 template <typename T>
 struct CombineSgdLauncher {
-  static hipError_t run(void* p, float self_w, const void* gathered,
-                        const float* w, int n_nbr, const void* grad, void* mom,
-                        float lr, float mu, float wd, float dampening,
+  using Acc = typename AccOf<T>::type;
+  static hipError_t run(void* p, double self_w, const void* gathered,
+                        const double* w, int n_nbr, const void* grad, void* mom,
+                        double lr, double mu, double wd, double dampening,
                         int nesterov, long numel, hipStream_t stream) {
     constexpr int V = vec_width<T>();
     if (n_nbr > kMaxNbrPerLaunch) {
@@ -402,10 +423,10 @@ struct CombineSgdLauncher {
       gathered = static_cast<const T*>(gathered) + static_cast<long>(overflow) * numel;
       w += overflow;
       n_nbr = kMaxNbrPerLaunch;
-      self_w = 1.0f;
+      self_w = 1.0;
     }
-    WeightsArg warg{};
-    for (int k = 0; k < n_nbr; ++k) warg.w[k] = w[k];
+    WeightsArgT<Acc> warg{};
+    for (int k = 0; k < n_nbr; ++k) warg.w[k] = static_cast<Acc>(w[k]);
     const bool vec = vec_ok<T>(p, numel) && vec_ok<T>(grad, numel) &&
                      (n_nbr == 0 || vec_ok<T>(gathered, numel)) &&
                      (mom == nullptr || vec_ok<T>(mom, numel));
@@ -413,10 +434,12 @@ struct CombineSgdLauncher {
     const bool has_mom = mom != nullptr;
 #define BF_LAUNCH_SGD(VV, MM)                                                  \
   hipLaunchKernelGGL((combine_sgd_k<T, VV, MM>), dim3(grid), dim3(kThreads), 0,\
-                     stream, static_cast<T*>(p), self_w,                       \
+                     stream, static_cast<T*>(p), static_cast<Acc>(self_w),     \
                      static_cast<const T*>(gathered), warg, n_nbr,             \
-                     static_cast<const T*>(grad), static_cast<T*>(mom), lr,    \
-                     mu, wd, dampening, nesterov, numel)
+                     static_cast<const T*>(grad), static_cast<T*>(mom),        \
+                     static_cast<Acc>(lr), static_cast<Acc>(mu),               \
+                     static_cast<Acc>(wd), static_cast<Acc>(dampening),        \
+                     nesterov, numel)
     if (vec) {
       if (has_mom)
         BF_LAUNCH_SGD(V, true);
@@ -435,11 +458,13 @@ struct CombineSgdLauncher {
 
 template <typename T>
 struct CombineAdamLauncher {
-  static hipError_t run(void* p, float self_w, const void* gathered,
-                        const float* w, int n_nbr, const void* grad,
-                        float* exp_avg, float* exp_avg_sq, float lr, float beta1,
-                        float beta2, float eps, float wd, float bias1,
-                        float bias2, long numel, hipStream_t stream) {
+  using Acc = typename AccOf<T>::type;
+  static hipError_t run(void* p, double self_w, const void* gathered,
+                        const double* w, int n_nbr, const void* grad,
+                        float* exp_avg, float* exp_avg_sq, double lr,
+                        double beta1, double beta2, double eps, double wd,
+                        double bias1, double bias2, long numel,
+                        hipStream_t stream) {
     if (n_nbr > kMaxNbrPerLaunch) {
       const int overflow = n_nbr - kMaxNbrPerLaunch;
       BF_CHECK_HIP((CombineLauncher<T>::run(p, p, self_w, gathered, w, overflow,
@@ -447,16 +472,19 @@ struct CombineAdamLauncher {
       gathered = static_cast<const T*>(gathered) + static_cast<long>(overflow) * numel;
       w += overflow;
       n_nbr = kMaxNbrPerLaunch;
-      self_w = 1.0f;
+      self_w = 1.0;
     }
-    WeightsArg warg{};
-    for (int k = 0; k < n_nbr; ++k) warg.w[k] = w[k];
+    WeightsArgT<Acc> warg{};
+    for (int k = 0; k < n_nbr; ++k) warg.w[k] = static_cast<Acc>(w[k]);
     const int grid = grid_for(numel);
     hipLaunchKernelGGL((combine_adam_k<T, 1>), dim3(grid), dim3(kThreads), 0,
-                       stream, static_cast<T*>(p), self_w,
+                       stream, static_cast<T*>(p), static_cast<Acc>(self_w),
                        static_cast<const T*>(gathered), warg, n_nbr,
-                       static_cast<const T*>(grad), exp_avg, exp_avg_sq, lr,
-                       beta1, beta2, eps, wd, bias1, bias2, numel);
+                       static_cast<const T*>(grad), exp_avg, exp_avg_sq,
+                       static_cast<Acc>(lr), static_cast<Acc>(beta1),
+                       static_cast<Acc>(beta2), static_cast<Acc>(eps),
+                       static_cast<Acc>(wd), static_cast<Acc>(bias1),
+                       static_cast<Acc>(bias2), numel);
     return hipGetLastError();
   }
 };
@@ -469,39 +497,39 @@ struct CombineAdamLauncher {
 
 extern "C" {
 
-hipError_t bf_weighted_combine(void* out, const void* self, float self_w,
-                               const void* gathered, const float* w, int n_nbr,
+hipError_t bf_weighted_combine(void* out, const void* self, double self_w,
+                               const void* gathered, const double* w, int n_nbr,
                                long numel, int dtype, hipStream_t stream) {
   return dispatch_dtype<CombineLauncher>(dtype, out, self, self_w, gathered, w,
                                          n_nbr, numel, stream);
 }
 
-hipError_t bf_scale_put(void* dst, const void* src, float w, long numel,
+hipError_t bf_scale_put(void* dst, const void* src, double w, long numel,
                         int dtype, bool accum, hipStream_t stream) {
   return dispatch_dtype<ScalePutLauncher>(dtype, dst, src, w, numel, accum,
                                           stream);
 }
 
-hipError_t bf_scale_inplace(void* buf, float f, long numel, int dtype,
+hipError_t bf_scale_inplace(void* buf, double f, long numel, int dtype,
                             hipStream_t stream) {
   return dispatch_dtype<ScaleInplaceLauncher>(dtype, buf, f, numel, stream);
 }
 
-hipError_t bf_combine_sgd(void* p, float self_w, const void* gathered,
-                          const float* w, int n_nbr, const void* grad,
-                          void* mom, float lr, float mu, float wd,
-                          float dampening, int nesterov, long numel, int dtype,
+hipError_t bf_combine_sgd(void* p, double self_w, const void* gathered,
+                          const double* w, int n_nbr, const void* grad,
+                          void* mom, double lr, double mu, double wd,
+                          double dampening, int nesterov, long numel, int dtype,
                           hipStream_t stream) {
   return dispatch_dtype<CombineSgdLauncher>(dtype, p, self_w, gathered, w,
                                             n_nbr, grad, mom, lr, mu, wd,
                                             dampening, nesterov, numel, stream);
 }
 
-hipError_t bf_combine_adam(void* p, float self_w, const void* gathered,
-                           const float* w, int n_nbr, const void* grad,
-                           float* exp_avg, float* exp_avg_sq, float lr,
-                           float beta1, float beta2, float eps, float wd,
-                           float bias1, float bias2, long numel, int dtype,
+hipError_t bf_combine_adam(void* p, double self_w, const void* gathered,
+                           const double* w, int n_nbr, const void* grad,
+                           float* exp_avg, float* exp_avg_sq, double lr,
+                           double beta1, double beta2, double eps, double wd,
+                           double bias1, double bias2, long numel, int dtype,
                            hipStream_t stream) {
   return dispatch_dtype<CombineAdamLauncher>(dtype, p, self_w, gathered, w,
                                              n_nbr, grad, exp_avg, exp_avg_sq,
