@@ -12,12 +12,23 @@ background-thread status slot:
   lazily on first synchronize/wait.
 """
 
+import os
 import threading
+import time
 from typing import Callable, List, Optional
 
 import torch
 
+from bluefog_amd.utils.logging import get_logger
 from bluefog_amd.utils.timeline import timeline
+
+# Stalled-op detection (reference analog: operations.cc:388-433 — the
+# coordinator reports tensors stuck in negotiation for >60 s together with
+# the ranks that never submitted them). Here a watchdog thread scans the
+# outstanding-handle table; a handle alive past the threshold is reported
+# once, with the op name, so a rank whose peers never posted the matching
+# send/recv is diagnosable. BLUEFOG_STALL_WARNING_TIME seconds, 0 disables.
+_STALL_WARNING_TIME = float(os.environ.get("BLUEFOG_STALL_WARNING_TIME", "60"))
 
 
 class OpHandle:
@@ -86,6 +97,44 @@ class HandleManager:
         self._next = 0
         self._handles = {}
         self._outstanding_names = set()
+        self._birth = {}
+        self._stall_reported = set()
+        self._watchdog: Optional[threading.Thread] = None
+
+    def _ensure_watchdog(self) -> None:
+        # started lazily with the first handle; daemon thread, wakes every
+        # few seconds — zero cost on the op hot path
+        if self._watchdog is not None or _STALL_WARNING_TIME <= 0:
+            return
+        t = threading.Thread(target=self._watch, daemon=True, name="bf-stall-watchdog")
+        self._watchdog = t
+        t.start()
+
+    def _watch(self) -> None:
+        period = min(10.0, max(1.0, _STALL_WARNING_TIME / 6.0))
+        while True:
+            time.sleep(period)
+            now = time.monotonic()
+            with self._lock:
+                stalled = [
+                    (hid, h.name, now - self._birth[hid])
+                    for hid, h in self._handles.items()
+                    if now - self._birth.get(hid, now) > _STALL_WARNING_TIME
+                    and hid not in self._stall_reported
+                    and not h.done
+                ]
+                for hid, _, _ in stalled:
+                    self._stall_reported.add(hid)
+            for hid, name, age in stalled:
+                if not self._handles.get(hid, OpHandle(-1, "")).poll():
+                    get_logger().warning(
+                        "op %r (handle %d) has not completed for %.0f s — "
+                        "one or more peer ranks likely never submitted the "
+                        "matching operation (stalled collective/p2p)",
+                        name,
+                        hid,
+                        age,
+                    )
 
     def allocate(self, name: str) -> OpHandle:
         with self._lock:
@@ -100,6 +149,8 @@ class HandleManager:
             h = OpHandle(hid, name)
             self._handles[hid] = h
             self._outstanding_names.add(name)
+            self._birth[hid] = time.monotonic()
+            self._ensure_watchdog()
             return h
 
     def get(self, hid: int) -> OpHandle:
@@ -112,6 +163,8 @@ class HandleManager:
     def release(self, hid: int) -> None:
         with self._lock:
             h = self._handles.pop(hid, None)
+            self._birth.pop(hid, None)
+            self._stall_reported.discard(hid)
             if h is not None:
                 self._outstanding_names.discard(h.name)
 

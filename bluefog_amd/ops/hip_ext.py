@@ -163,6 +163,8 @@ def weighted_combine_sgd(
     lr: float,
     momentum: float,
     weight_decay: float,
+    dampening: float = 0.0,
+    nesterov: bool = False,
 ) -> None:
     """param = combine(param, neighbors); then SGD(momentum) step in the same
     pass over HBM."""
@@ -177,6 +179,8 @@ def weighted_combine_sgd(
             float(lr),
             float(momentum),
             float(weight_decay),
+            float(dampening),
+            bool(nesterov),
         )
         return
     weighted_combine(param, param.clone(), self_weight, gathered, weights)
@@ -184,8 +188,8 @@ def weighted_combine_sgd(
     if weight_decay != 0:
         g = g.add(param, alpha=weight_decay)
     if momentum_buf is not None and momentum != 0:
-        momentum_buf.mul_(momentum).add_(g)
-        g = momentum_buf
+        momentum_buf.mul_(momentum).add_(g, alpha=1.0 - dampening)
+        g = grad.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
     param.add_(g, alpha=-lr)
 
 

@@ -151,3 +151,37 @@ def test_single_process_init():
         [sys.executable, "-c", code], capture_output=True, text=True, timeout=120
     )
     assert r.returncode == 0 and "OK" in r.stdout, r.stderr
+
+
+def test_stall_watchdog_warns(monkeypatch, caplog):
+    """Stalled-op detection (reference operations.cc:388-433): an op that
+    never completes is reported with its name after the threshold."""
+    import logging
+    import time as _time
+
+    from bluefog_amd.ops import handles as H
+
+    monkeypatch.setattr(H, "_STALL_WARNING_TIME", 0.2)
+    m = H.HandleManager()
+    h = m.allocate("neighbor_allreduce.stalled_param")
+
+    class _NeverDone:
+        def is_completed(self):
+            return False
+
+    h.works = [_NeverDone()]
+    from bluefog_amd.utils.logging import get_logger
+
+    get_logger().addHandler(caplog.handler)  # logger is non-propagating
+    with caplog.at_level(logging.WARNING, logger="bluefog_amd"):
+        deadline = _time.time() + 5.0
+        while _time.time() < deadline and not any(
+            "stalled" in r.message for r in caplog.records
+        ):
+            _time.sleep(0.1)
+    get_logger().removeHandler(caplog.handler)
+    m.release(h.id)
+    assert any(
+        "neighbor_allreduce.stalled_param" in r.message and "stalled" in r.message
+        for r in caplog.records
+    )
