@@ -282,9 +282,43 @@ __global__ __launch_bounds__(kThreads) void combine_adam_k(
     typename AccOf<T>::type bias1, typename AccOf<T>::type bias2,
     long numel) {
   using Acc = typename AccOf<T>::type;
+  const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  long t = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  for (; t < numel; t += stride) {
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> pv = *reinterpret_cast<const Pack<T, VEC>*>(p + base);
+    Acc acc[VEC];
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) acc[v] = self_w * AccOf<T>::to(pv.v[v]);
+    for (int k = 0; k < n_nbr; ++k) {
+      Pack<T, VEC> gv =
+          *reinterpret_cast<const Pack<T, VEC>*>(gathered + static_cast<long>(k) * numel + base);
+      const Acc wk = warg.w[k];
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) acc[v] += wk * AccOf<T>::to(gv.v[v]);
+    }
+    Pack<T, VEC> grv = *reinterpret_cast<const Pack<T, VEC>*>(grad + base);
+    // fp32 optimizer state: VEC floats may span several 16B packs; load as
+    // Pack<float,4> chunks so the compiler emits dwordx4 accesses
+    Pack<float, VEC> mv = *reinterpret_cast<const Pack<float, VEC>*>(exp_avg + base);
+    Pack<float, VEC> vv = *reinterpret_cast<const Pack<float, VEC>*>(exp_avg_sq + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      Acc g = AccOf<T>::to(grv.v[v]) + wd * acc[v];
+      Acc m = beta1 * static_cast<Acc>(mv.v[v]) + (static_cast<Acc>(1) - beta1) * g;
+      Acc vt = beta2 * static_cast<Acc>(vv.v[v]) + (static_cast<Acc>(1) - beta2) * g * g;
+      mv.v[v] = static_cast<float>(m);
+      vv.v[v] = static_cast<float>(vt);
+      const Acc denom = sqrt(vt / bias2) + eps;
+      pv.v[v] = AccOf<T>::from(acc[v] - lr * (m / bias1) / denom);
+    }
+    *reinterpret_cast<Pack<T, VEC>*>(p + base) = pv;
+    *reinterpret_cast<Pack<float, VEC>*>(exp_avg + base) = mv;
+    *reinterpret_cast<Pack<float, VEC>*>(exp_avg_sq + base) = vv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
     Acc acc = self_w * AccOf<T>::to(p[t]);
     for (int k = 0; k < n_nbr; ++k)
       acc += warg.w[k] * AccOf<T>::to(gathered[static_cast<long>(k) * numel + t]);
@@ -476,15 +510,26 @@ struct CombineAdamLauncher {
     }
     WeightsArgT<Acc> warg{};
     for (int k = 0; k < n_nbr; ++k) warg.w[k] = static_cast<Acc>(w[k]);
-    const int grid = grid_for(numel);
-    hipLaunchKernelGGL((combine_adam_k<T, 1>), dim3(grid), dim3(kThreads), 0,
-                       stream, static_cast<T*>(p), static_cast<Acc>(self_w),
-                       static_cast<const T*>(gathered), warg, n_nbr,
-                       static_cast<const T*>(grad), exp_avg, exp_avg_sq,
-                       static_cast<Acc>(lr), static_cast<Acc>(beta1),
-                       static_cast<Acc>(beta2), static_cast<Acc>(eps),
-                       static_cast<Acc>(wd), static_cast<Acc>(bias1),
-                       static_cast<Acc>(bias2), numel);
+    constexpr int V = vec_width<T>();
+    const bool vec = vec_ok<T>(p, numel) && vec_ok<T>(grad, numel) &&
+                     vec_ok<float>(exp_avg, numel) &&
+                     vec_ok<float>(exp_avg_sq, numel) &&
+                     (n_nbr == 0 || vec_ok<T>(gathered, numel));
+    const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
+#define BF_LAUNCH_ADAM(VV)                                                     \
+  hipLaunchKernelGGL((combine_adam_k<T, VV>), dim3(grid), dim3(kThreads), 0,   \
+                     stream, static_cast<T*>(p), static_cast<Acc>(self_w),     \
+                     static_cast<const T*>(gathered), warg, n_nbr,             \
+                     static_cast<const T*>(grad), exp_avg, exp_avg_sq,         \
+                     static_cast<Acc>(lr), static_cast<Acc>(beta1),            \
+                     static_cast<Acc>(beta2), static_cast<Acc>(eps),           \
+                     static_cast<Acc>(wd), static_cast<Acc>(bias1),            \
+                     static_cast<Acc>(bias2), numel)
+    if (vec)
+      BF_LAUNCH_ADAM(V);
+    else
+      BF_LAUNCH_ADAM(1);
+#undef BF_LAUNCH_ADAM
     return hipGetLastError();
   }
 };

@@ -1,0 +1,54 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""ibfrun interactive-cluster tests (reference analog: the ipyparallel
+ibfrun of bluefog/run/interactive_run.py)."""
+
+import os
+
+import pytest
+
+from bluefog_amd.run import interactive as ib
+
+
+@pytest.fixture
+def cluster(tmp_path, monkeypatch):
+    monkeypatch.setattr(ib, "_STATE_DIR", str(tmp_path))
+    state = ib.start_cluster(2, profile="test", extra_env={"BLUEFOG_LOG_LEVEL": "error"})
+    yield state
+    ib.stop_cluster(profile="test")
+
+
+def _init_and_allreduce():
+    import torch
+
+    import bluefog_amd as bf
+
+    bf.init()
+    t = torch.ones(4) * (bf.rank() + 1)
+    out = bf.allreduce(t, average=True, name="ib_t")
+    return float(out[0])
+
+
+def test_interactive_roundtrip(cluster):
+    c = ib.InteractiveClient(profile="test")
+    assert c.ping() == ["0", "1"]
+    # persistent namespace
+    c.run_code("x = int(__import__('os').environ['RANK']) * 10")
+    assert c.pull("x") == [0, 10]
+    # real collective over gloo inside the workers
+    outs = c.run(_init_and_allreduce)
+    assert outs == [1.5, 1.5]
+    c.close()
+
+
+def _boom():
+    raise RuntimeError("worker exploded")
+
+
+def test_interactive_error_propagates(cluster):
+    c = ib.InteractiveClient(profile="test")
+
+    with pytest.raises(ib.ClusterError, match="worker exploded"):
+        c.run(_boom)
+    # cluster still usable afterwards
+    assert c.ping() == ["0", "1"]
+    c.close()
