@@ -9,6 +9,12 @@ drives every rank to the global mean.
     ./bfrun -np 2 python examples/pytorch_average_consensus.py --asynchronous-mode
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 
 import torch

@@ -9,6 +9,12 @@ batches, per-rank numbers summed by allreduce).
         --model resnet50 --batch-size 64 --dist-optimizer neighbor_allreduce
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import time
 

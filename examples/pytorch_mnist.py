@@ -8,6 +8,12 @@ absent; pass --data-dir to use downloaded MNIST tensors if available.
     ./bfrun -np 4 python examples/pytorch_mnist.py --epochs 2
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import os
 

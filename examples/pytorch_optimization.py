@@ -8,6 +8,12 @@ are the published ones, implemented here from their update equations).
     ./bfrun -np 4 python examples/pytorch_optimization.py --method diffusion
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 
 import torch

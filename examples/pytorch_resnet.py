@@ -14,6 +14,12 @@ tensor files to train on real data.
     ./bfrun -np 8 python examples/pytorch_resnet.py --dist-optimizer win_put
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import math
 import os
