@@ -33,6 +33,10 @@ hipError_t bf_combine_adam(void* p, double self_w, const void* gathered,
                            double beta1, double beta2, double eps, double wd,
                            double bias1, double bias2, long numel, int dtype,
                            hipStream_t stream);
+hipError_t bf_add_relu_fwd(void* out, const void* a, const void* b, long numel,
+                           int dtype, hipStream_t stream);
+hipError_t bf_relu_bwd_mask(void* gin, const void* g, const void* out,
+                            long numel, int dtype, hipStream_t stream);
 }
 
 namespace {
@@ -151,6 +155,35 @@ void weighted_combine_adam(at::Tensor param, double self_weight,
       "weighted_combine_adam");
 }
 
+void add_relu_fwd(at::Tensor out, at::Tensor a, at::Tensor b) {
+  // elementwise over identically-laid-out dense tensors (plain contiguous
+  // or channels_last both qualify)
+  TORCH_CHECK(out.is_non_overlapping_and_dense() &&
+                  a.is_non_overlapping_and_dense() &&
+                  b.is_non_overlapping_and_dense() &&
+                  a.strides() == b.strides() && out.strides() == a.strides(),
+              "add_relu needs dense tensors with identical layout");
+  TORCH_CHECK(a.numel() == b.numel() && out.numel() == a.numel(), "numel mismatch");
+  TORCH_CHECK(a.scalar_type() == b.scalar_type() &&
+                  out.scalar_type() == a.scalar_type(),
+              "dtype mismatch");
+  check_hip(bf_add_relu_fwd(out.data_ptr(), a.data_ptr(), b.data_ptr(),
+                            a.numel(), dtype_code(a), current_stream()),
+            "add_relu_fwd");
+}
+
+void relu_bwd_mask(at::Tensor gin, at::Tensor g, at::Tensor out) {
+  TORCH_CHECK(gin.is_non_overlapping_and_dense() &&
+                  g.is_non_overlapping_and_dense() &&
+                  out.is_non_overlapping_and_dense() &&
+                  g.strides() == out.strides() && gin.strides() == g.strides(),
+              "relu_bwd_mask needs dense tensors with identical layout");
+  TORCH_CHECK(g.numel() == out.numel() && gin.numel() == g.numel(), "numel mismatch");
+  check_hip(bf_relu_bwd_mask(gin.data_ptr(), g.data_ptr(), out.data_ptr(),
+                             g.numel(), dtype_code(g), current_stream()),
+            "relu_bwd_mask");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -164,4 +197,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused neighbor-average + SGD(momentum) over a flat bucket");
   m.def("weighted_combine_adam", &weighted_combine_adam,
         "fused neighbor-average + Adam over a flat bucket");
+  m.def("add_relu_fwd", &add_relu_fwd, "out = max(a+b, 0)");
+  m.def("relu_bwd_mask", &relu_bwd_mask, "gin = out>0 ? g : 0");
 }

@@ -9,6 +9,8 @@ from typing import List, Type, Union
 import torch
 import torch.nn as nn
 
+from bluefog_amd.ops.fused_modules import add_relu
+
 __all__ = ["ResNet", "resnet18", "resnet34", "resnet50", "resnet101", "resnet152"]
 
 
@@ -30,7 +32,7 @@ class BasicBlock(nn.Module):
         out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        return add_relu(out, identity)
 
 
 class Bottleneck(nn.Module):
@@ -54,7 +56,7 @@ class Bottleneck(nn.Module):
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        return add_relu(out, identity)
 
 
 class ResNet(nn.Module):

@@ -215,3 +215,27 @@ def test_resnet_step_gpu():
         opt.step()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16, torch.float16])
+@pytest.mark.parametrize("channels_last", [False, True])
+def test_add_relu_gpu_matches_eager(dev, dtype, channels_last):
+    from bluefog_amd.ops.fused_modules import add_relu
+
+    torch.manual_seed(0)
+    shape = (8, 32, 14, 14)
+    mf = torch.channels_last if channels_last else torch.contiguous_format
+    a = torch.randn(shape, device=dev).to(dtype).to(memory_format=mf).requires_grad_()
+    b = torch.randn(shape, device=dev).to(dtype).to(memory_format=mf).requires_grad_()
+    a2 = a.detach().clone().requires_grad_()
+    b2 = b.detach().clone().requires_grad_()
+    out = add_relu(a, b)
+    ref = torch.relu(a2 + b2)
+    torch.cuda.synchronize()
+    assert torch.equal(out, ref)
+    g = torch.randn(shape, device=dev).to(dtype).to(memory_format=mf)
+    out.backward(g)
+    ref.backward(g)
+    torch.cuda.synchronize()
+    assert torch.equal(a.grad, a2.grad)
+    assert torch.equal(b.grad, b2.grad)
