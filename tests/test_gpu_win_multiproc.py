@@ -39,15 +39,17 @@ h = bf.win_put_nonblocking(t, "w")
 bf.win_wait(h)
 bf.barrier()
 torch.cuda.synchronize()
+# versions count un-acknowledged puts; the peer's put is pending
+ver = bf.get_win_version("w")
+assert all(v >= 1 for v in ver.values()), ver
 out = bf.win_update("w")
 # ring(2): one in-neighbor; uniform weights 1/2 each -> (1+2)/2 = 1.5
 expected = ( (rank + 1.0) + (2.0 - rank) ) / 2.0
 assert torch.allclose(out, torch.full_like(out, expected)), (
     rank, float(out.mean()))
-
-# versions bumped by the put
+# win_update acknowledged the put
 ver = bf.get_win_version("w")
-assert all(v >= 1 for v in ver.values()), ver
+assert all(v == 0 for v in ver.values()), ver
 
 # ---- win_accumulate ------------------------------------------------------
 t2 = torch.ones(1000, device=dev) * (rank + 1.0)
