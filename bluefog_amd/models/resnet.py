@@ -65,6 +65,7 @@ class ResNet(nn.Module):
         block: Type[Union[BasicBlock, Bottleneck]],
         layers: List[int],
         num_classes: int = 1000,
+        zero_init_residual: bool = False,
     ):
         super().__init__()
         self.in_planes = 64
@@ -84,6 +85,14 @@ class ResNet(nn.Module):
             elif isinstance(m, nn.BatchNorm2d):
                 nn.init.constant_(m.weight, 1)
                 nn.init.constant_(m.bias, 0)
+        if zero_init_residual:
+            # start each residual branch at identity: stabilizes early
+            # training (He et al. "Bag of Tricks", standard option)
+            for m in self.modules():
+                if isinstance(m, Bottleneck):
+                    nn.init.zeros_(m.bn3.weight)
+                elif isinstance(m, BasicBlock):
+                    nn.init.zeros_(m.bn2.weight)
 
     def _make_layer(self, block, planes, blocks, stride=1):
         downsample = None
@@ -105,21 +114,21 @@ class ResNet(nn.Module):
         return self.fc(x)
 
 
-def resnet18(num_classes=1000):
-    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes)
+def resnet18(num_classes=1000, zero_init_residual=False):
+    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes, zero_init_residual)
 
 
-def resnet34(num_classes=1000):
-    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes)
+def resnet34(num_classes=1000, zero_init_residual=False):
+    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes, zero_init_residual)
 
 
-def resnet50(num_classes=1000):
-    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes)
+def resnet50(num_classes=1000, zero_init_residual=False):
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, zero_init_residual)
 
 
-def resnet101(num_classes=1000):
-    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes)
+def resnet101(num_classes=1000, zero_init_residual=False):
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, zero_init_residual)
 
 
-def resnet152(num_classes=1000):
-    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes)
+def resnet152(num_classes=1000, zero_init_residual=False):
+    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes, zero_init_residual)

@@ -38,7 +38,7 @@ parser.add_argument("--data-dir", default=None, help="dir with train.pt/val.pt; 
 parser.add_argument("--batch-size", type=int, default=32)
 parser.add_argument("--val-batch-size", type=int, default=32)
 parser.add_argument("--epochs", type=int, default=3)
-parser.add_argument("--base-lr", type=float, default=0.0125, help="per-worker lr")
+parser.add_argument("--base-lr", type=float, default=0.003, help="per-worker lr")
 parser.add_argument("--warmup-epochs", type=float, default=1)
 parser.add_argument("--momentum", type=float, default=0.9)
 parser.add_argument("--wd", type=float, default=5e-5)
@@ -116,7 +116,7 @@ def main():
 
     model = {"resnet18": resnet18, "resnet50": resnet50, "resnet101": resnet101}[
         args.model
-    ](num_classes=args.num_classes).to(device)
+    ](num_classes=args.num_classes, zero_init_residual=True).to(device)
 
     optimizer = torch.optim.SGD(
         model.parameters(),
