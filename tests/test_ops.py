@@ -258,6 +258,87 @@ def w_topo_check_mismatch():
         pass
 
 
+
+
+def w_consensus_convergence():
+    """Iterated static-exp2 neighbor averaging drives every rank to the
+    global mean — the framework's core semantic, end to end (reference
+    examples/pytorch_average_consensus.py as a test)."""
+    import bluefog_amd as bf
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    bf.set_topology(bf.ExponentialTwoGraph(size))
+    torch.manual_seed(100 + rank)
+    x = torch.randn(64, dtype=torch.float64)
+    mean = bf.allreduce(x, average=True, name="true_mean")
+    for _ in range(40):
+        x = bf.neighbor_allreduce(x)
+    assert torch.allclose(x, mean, atol=1e-8), float((x - mean).abs().max())
+
+
+def w_neighbor_allreduce_half():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    for dtype in HALF_DTYPES:
+        t = torch.ones(33, dtype=dtype) * rank
+        out = bf.neighbor_allreduce(t)
+        left, right = (rank - 1) % size, (rank + 1) % size
+        nbrs = [left] if size == 2 else [left, right]
+        expected = (rank + sum(nbrs)) / (len(nbrs) + 1)
+        assert out.dtype == dtype
+        assert torch.allclose(
+            out.float(), torch.full((33,), expected), atol=2e-2
+        ), (dtype, rank, out[0].item(), expected)
+
+
+def w_neighbor_allreduce_dims():
+    bf = _init_ring()
+    rank, size = bf.rank(), bf.size()
+    left, right = (rank - 1) % size, (rank + 1) % size
+    nbrs = [left] if size == 2 else [left, right]
+    expected = (rank + sum(nbrs)) / (len(nbrs) + 1)
+    for shape in [(1,), (23,), (4, 5), (2, 3, 4), (2, 2, 2, 2)]:
+        t = torch.ones(shape, dtype=torch.float32) * rank
+        out = bf.neighbor_allreduce(t)
+        assert out.shape == t.shape
+        assert torch.allclose(out, torch.full(shape, expected), atol=1e-6)
+
+
+def w_inner_outer_dynamic():
+    """Comm-exercise the InnerOuterRing/Expo2 dynamic generators (reference
+    torch_ops_test dynamic move patterns): every iteration the claimed
+    send/recv sets must be mutually consistent and produce exact averages."""
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    bf.set_topology(bf.ExponentialTwoGraph(size))
+    for gen_fn in (
+        tu.GetInnerOuterRingDynamicSendRecvRanks,
+        tu.GetInnerOuterExpo2DynamicSendRecvRanks,
+    ):
+        gen = gen_fn(size, local_size=4, self_rank=rank)
+        for it in range(6):
+            send_ranks, recv_ranks = next(gen)
+            w = 1.0 / (len(recv_ranks) + 1)
+            t = torch.ones(7, dtype=torch.float64) * (rank + 1)
+            out = bf.neighbor_allreduce(
+                t,
+                self_weight=w,
+                src_weights={r: w for r in recv_ranks},
+                dst_weights=send_ranks,
+                enable_topo_check=True,
+            )
+            expected = w * (rank + 1) + sum(w * (r + 1) for r in recv_ranks)
+            assert torch.allclose(out, torch.full_like(t, expected)), (
+                gen_fn.__name__,
+                it,
+                rank,
+            )
+
+
 # --------------------------------------------------------------------------
 # pytest entry points
 # --------------------------------------------------------------------------
@@ -315,3 +396,18 @@ def test_barrier():
 
 def test_topo_check_mismatch():
     run_dist(w_topo_check_mismatch, 2)
+
+def test_consensus_convergence():
+    run_dist(w_consensus_convergence, 4)
+
+
+def test_neighbor_allreduce_half():
+    run_dist(w_neighbor_allreduce_half, 2)
+
+
+def test_neighbor_allreduce_dims():
+    run_dist(w_neighbor_allreduce_dims, 2)
+
+
+def test_inner_outer_dynamic():
+    run_dist(w_inner_outer_dynamic, 8)
