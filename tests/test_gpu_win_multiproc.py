@@ -122,3 +122,83 @@ def test_gpu_ipc_windows_two_ranks_one_device():
     for rank, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0, f"rank {rank} failed:\n{out}"
         assert f"MULTIPROC_WIN_OK rank={rank}" in out, out
+
+
+_OPT_WORKER = r"""
+import os, sys
+sys.path.insert(0, os.environ["BF_ROOT"])
+import torch
+import torch.nn as nn
+import bluefog_amd as bf
+
+bf.init()
+rank, size = bf.rank(), bf.size()
+torch.cuda.set_device(0)
+dev = torch.device("cuda:0")
+bf.set_topology(bf.RingGraph(size))
+
+torch.manual_seed(3)
+model = nn.Sequential(nn.Linear(32, 64), nn.ReLU(), nn.Linear(64, 1)).to(dev)
+opt = bf.DistributedWinPutOptimizer(
+    torch.optim.SGD(model.parameters(), lr=0.03), model=model
+)
+bf.broadcast_parameters(model.state_dict(), root_rank=0)
+
+# synthetic linear regression, different shard per rank
+g = torch.Generator().manual_seed(50 + rank)
+X = torch.randn(256, 32, generator=g).to(dev)
+w_true = torch.arange(32, dtype=torch.float32).to(dev) / 32.0
+Y = (X @ w_true).unsqueeze(1) + 0.01 * torch.randn(256, 1, generator=g).to(dev)
+
+loss0 = None
+for step in range(30):
+    opt.zero_grad()
+    loss = ((model(X) - Y) ** 2).mean()
+    loss.backward()
+    opt.step()
+    if step == 0:
+        loss0 = float(loss)
+final = float(loss)
+assert final < loss0 * 0.2, (loss0, final)
+opt.unregister_window()
+print(f"WINPUT_OPT_OK rank={rank} loss {loss0:.4f}->{final:.4f}", flush=True)
+"""
+
+
+def test_gpu_win_put_optimizer_two_ranks_one_device():
+    """End-to-end async-gossip training (DistributedWinPutOptimizer, the
+    BASELINE config-4 path) with the HIP-IPC window data plane: two ranks
+    share cuda:0, parameters gossip through win_put + win_update."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    port = _free_port()
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update(
+            BF_ROOT=_ROOT,
+            RANK=str(rank),
+            LOCAL_RANK=str(rank),
+            WORLD_SIZE="2",
+            MASTER_ADDR="127.0.0.1",
+            MASTER_PORT=str(port),
+            BLUEFOG_BACKEND="gloo",
+            HSA_ENABLE_IPC_MODE_LEGACY="0",
+        )
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, "-c", _OPT_WORKER],
+                env=env,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT,
+            )
+        )
+    for rank, p in enumerate(procs):
+        try:
+            out, _ = p.communicate(timeout=300)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            out, _ = p.communicate()
+        text = out.decode(errors="replace")
+        assert p.returncode == 0, f"rank {rank} failed:\n{text}"
+        assert f"WINPUT_OPT_OK rank={rank}" in text, text
