@@ -279,3 +279,47 @@ def test_associated_p():
 
 def test_push_sum_consistency():
     run_dist(w_push_sum_consistency, 4, timeout=300)
+
+
+def w_win_mutex_stress():
+    """Atomicity stress: rank 0 fires many win_accumulates while rank 1
+    concurrently runs win_update_then_collect (which resets the buffer).
+    Under the distributed mutex no update may be lost: at the end rank 1's
+    window tensor must hold exactly the sum of every accumulate."""
+    import bluefog_amd as bf
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    bf.set_topology(bf.RingGraph(size))
+    n_puts = 40
+    t = torch.zeros(257)
+    bf.win_create(t, "stress", zero_init=True)
+    bf.barrier()
+    if rank == 0:
+        ones = torch.ones(257)
+        for _ in range(n_puts):
+            bf.win_accumulate(ones, "stress", dst_weights={1: 1.0},
+                              require_mutex=True)
+    else:
+        import time
+
+        # collect concurrently with the accumulates; every collected value
+        # folds into the window tensor, so it must converge to exactly
+        # n_puts — a lost update (accumulate racing the reset) would
+        # undershoot forever, a torn read would overshoot
+        deadline = time.time() + 120
+        out = t
+        while time.time() < deadline and out[0] < n_puts:
+            out = bf.win_update_then_collect("stress")
+            time.sleep(0.001)
+        assert torch.allclose(out, torch.full((257,), float(n_puts))), (
+            float(out.min()),
+            float(out.max()),
+            n_puts,
+        )
+    bf.barrier()
+    bf.win_free()
+
+
+def test_win_mutex_stress():
+    run_dist(w_win_mutex_stress, 2, timeout=180.0)
