@@ -876,13 +876,26 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
         self._parameter_names = {
             v: self.window_prefix + k for k, v in sorted(named_parameters)
         }
+        # MI355X design note: the reference registers ONE WINDOW PER
+        # PARAMETER (optimizers.py:933-944) — ~161 windows for ResNet50,
+        # each paying per-iteration host work (executor job, kernel launch
+        # per destination, store round-trips for versions/mutex). Here the
+        # parameters are flattened into a few large flat buckets (the same
+        # _FlatBuckets the AWC fast path uses) and each BUCKET gets one
+        # window: a handful of big xGMI transfers and O(buckets) host work
+        # per iteration instead of O(params).
+        self._buckets = _FlatBuckets(named_parameters, fusion_threshold_bytes())
+        self._bucket_names = {
+            id(b["flat"]): f"{self.window_prefix}winbucket.{i}"
+            for i, b in enumerate(self._buckets.buckets)
+        }
         self._handles = {}
         self._synchronized = False
         self._should_synchronize = True
         self._error_encountered = False
         self._num_steps_per_communication = num_steps_per_communication
         self._delay = {
-            v: num_steps_per_communication for _, v in sorted(named_parameters)
+            name: num_steps_per_communication for name in self._bucket_names.values()
         }
         self._timeline_hook_handles = []
         self._use_timeline = False
@@ -904,32 +917,31 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
             hook = self._make_get_hook() if self._pull_style else self._make_put_hook()
             model.register_forward_hook(hook)
 
-    def _fire(self, p, name):
+    def _fire(self, flat, name):
         bf = _bf()
         if self._pull_style:
             return bf.win_get_nonblocking(
                 name=name, src_weights=self.src_weights, require_mutex=True
             )
         return bf.win_put_nonblocking(
-            tensor=p.data, name=name, dst_weights=self.dst_weights, require_mutex=False
+            tensor=flat, name=name, dst_weights=self.dst_weights, require_mutex=False
         )
 
     def _make_hook_impl(self):
         def hook(model, *unused):
             if not model.training:
                 return
-            for p in (
-                q for group in self.param_groups for q in group["params"] if q.requires_grad
-            ):
-                if p in self._handles:
+            for b in self._buckets.buckets:
+                name = self._bucket_names[id(b["flat"])]
+                if name in self._handles:
                     continue
-                if self._delay[p] <= 0:
+                if self._delay[name] <= 0:
                     if not self._error_encountered:
                         warnings.warn(_warning_message_num_step_per_communication)
                         self._error_encountered = True
-                self._delay[p] -= 1
-                if self._delay[p] == 0:
-                    self._handles[p] = self._fire(p, self._parameter_names.get(p))
+                self._delay[name] -= 1
+                if self._delay[name] == 0:
+                    self._handles[name] = self._fire(b["flat"], name)
 
         return hook
 
@@ -941,23 +953,19 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
 
     def _register_window(self):
         bf = _bf()
-        for param_group in self.param_groups:
-            for p in param_group["params"]:
-                name = self._parameter_names.get(p)
-                if name is None:
-                    raise KeyError("Cannot find parameter in _parameter_names")
-                if not bf.win_create(p.data, name):
-                    raise ValueError(f"Cannot allocate window for parameter {name}")
+        for b in self._buckets.buckets:
+            name = self._bucket_names[id(b["flat"])]
+            if not bf.win_create(b["flat"], name):
+                raise ValueError(f"Cannot allocate window for bucket {name}")
 
     def unregister_window(self):
         bf = _bf()
         if bf.size() <= 1:
             return
-        for param_group in self.param_groups:
-            for p in param_group["params"]:
-                name = self._parameter_names.get(p)
-                if name in bf.get_current_created_window_names():
-                    bf.win_free(name)
+        for b in self._buckets.buckets:
+            name = self._bucket_names[id(b["flat"])]
+            if name in bf.get_current_created_window_names():
+                bf.win_free(name)
 
     def turn_on_timeline(self):
         self._use_timeline = True
@@ -976,11 +984,12 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
     def synchronize(self):
         bf = _bf()
         with torch.no_grad():
-            for p, handle in self._handles.items():
+            for name, handle in self._handles.items():
                 bf.win_wait(handle)
-                name = self._parameter_names.get(p)
-                self._delay[p] = self._num_steps_per_communication
-                p.set_(bf.win_update(name=name, require_mutex=True))
+                self._delay[name] = self._num_steps_per_communication
+                # win_update folds the neighbor buffers into the bucket's
+                # flat tensor IN PLACE; the parameters are views of it
+                bf.win_update(name=name, require_mutex=True)
         self._handles.clear()
         self._synchronized = True
 
@@ -1017,6 +1026,14 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
         named_parameters, models = _check_named_parameters(self, model)
         self._models = models
         self._parameter_names = {v: k for k, v in sorted(named_parameters)}
+        # bucketed windows, same rationale as _DistributedWinOptimizer: one
+        # extended window (bucket + trailing p scalar) per flat bucket
+        # instead of the reference's one per parameter
+        self._buckets = _FlatBuckets(named_parameters, fusion_threshold_bytes())
+        self._bucket_names = {
+            id(b["flat"]): f"pushsum.winbucket.{i}"
+            for i, b in enumerate(self._buckets.buckets)
+        }
         self._handles = {}
         self._named_ps_weights = {}
         self._named_extension_parameters = {}
@@ -1025,7 +1042,7 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
         self._error_encountered = False
         self._num_steps_per_communication = num_steps_per_communication
         self._delay = {
-            v: num_steps_per_communication for _, v in sorted(named_parameters)
+            name: num_steps_per_communication for name in self._bucket_names.values()
         }
         self._timeline_hook_handles = []
         self._use_timeline = False
@@ -1039,17 +1056,15 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
     @torch.no_grad()
     def _register_window(self):
         bf = _bf()
-        for param_group in self.param_groups:
-            for p in param_group["params"]:
-                name = self._parameter_names.get(p)
-                if name is None:
-                    raise KeyError("Cannot find parameter in _parameter_names")
-                ps_weights = torch.Tensor([1.0]).to(p.data.dtype).to(p.data.device)
-                self._named_ps_weights[name] = ps_weights
-                extended = torch.cat((p.data.view(-1), ps_weights), 0)
-                self._named_extension_parameters[name] = extended
-                if not bf.win_create(extended, name, zero_init=True):
-                    raise ValueError(f"Cannot allocate window for parameter {name}")
+        for b in self._buckets.buckets:
+            flat = b["flat"]
+            name = self._bucket_names[id(flat)]
+            ps_weights = torch.Tensor([1.0]).to(flat.dtype).to(flat.device)
+            self._named_ps_weights[name] = ps_weights
+            extended = torch.cat((flat, ps_weights), 0)
+            self._named_extension_parameters[name] = extended
+            if not bf.win_create(extended, name, zero_init=True):
+                raise ValueError(f"Cannot allocate window for bucket {name}")
 
     def _register_hooks(self):
         for model in self._models:
@@ -1060,23 +1075,21 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
             if not model.training:
                 return
             bf = _bf()
-            for p in (
-                q for group in self.param_groups for q in group["params"] if q.requires_grad
-            ):
-                if p in self._handles:
+            for b in self._buckets.buckets:
+                name = self._bucket_names[id(b["flat"])]
+                if name in self._handles:
                     continue
-                if self._delay[p] <= 0:
+                if self._delay[name] <= 0:
                     if not self._error_encountered:
                         warnings.warn(_warning_message_num_step_per_communication)
                         self._error_encountered = True
-                self._delay[p] -= 1
-                if self._delay[p] == 0:
-                    name = self._parameter_names.get(p)
+                self._delay[name] -= 1
+                if self._delay[name] == 0:
                     with torch.no_grad():
                         ext = self._named_extension_parameters[name]
-                        ext[:-1].copy_(p.data.view(-1))
+                        ext[:-1].copy_(b["flat"])
                         ext[-1] = self._named_ps_weights[name].item()
-                    self._handles[p] = bf.win_accumulate_nonblocking(
+                    self._handles[name] = bf.win_accumulate_nonblocking(
                         tensor=ext,
                         name=name,
                         dst_weights=self.dst_weights,
@@ -1101,18 +1114,18 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
 
     def synchronize(self):
         bf = _bf()
+        flats = {self._bucket_names[id(b["flat"])]: b["flat"] for b in self._buckets.buckets}
         with torch.no_grad():
-            for p, handle in self._handles.items():
+            for name, handle in self._handles.items():
                 bf.win_wait(handle)
-                name = self._parameter_names.get(p)
-                self._delay[p] = self._num_steps_per_communication
+                self._delay[name] = self._num_steps_per_communication
                 ext = self._named_extension_parameters[name]
                 # keep self's share, fold in neighbors' accumulations
                 ext.mul_(self.self_weight)
                 ext = bf.win_update_then_collect(name=name)
                 self._named_ps_weights[name].fill_(ext[-1].item())
-                corrected = (ext[:-1] / ext[-1]).reshape(p.shape)
-                p.set_(corrected)
+                # corrected bucket written back in place; params are views
+                flats[name].copy_(ext[:-1] / ext[-1])
         self._handles.clear()
         self._synchronized = True
 
