@@ -192,15 +192,19 @@ class _FlatBuckets:
 
 class _DistributedOptimizer(torch.optim.Optimizer):
     """Synchronous gradient averaging over all ranks (reference
-    optimizers.py:166-294): per-parameter nonblocking allreduce fired from a
-    post-grad-accumulation hook, synchronized at step()."""
+    optimizers.py:166-294). The reference fires one nonblocking allreduce
+    per parameter and leans on the coordinator's request fusion to batch
+    them; here gradients live in persistent flat buckets (views, like the
+    AWC fast path) and each bucket's allreduce fires from the backward
+    hooks as soon as every gradient in it has accumulated — DDP-style
+    bucketing sized for xGMI, still overlapped with the rest of backward."""
 
     def __init__(self, params, model, backward_passes_per_step=1):
         super(self.__class__, self).__init__(params)
         named_parameters, models = _check_named_parameters(self, model)
         self._models = models
         self._parameter_names = {v: k for k, v in sorted(named_parameters)}
-        self._handles = {}
+        self._handles = {}  # bucket-name -> handle
         self._synchronized = False
         self._should_synchronize = True
         self._error_encountered = False
@@ -208,6 +212,25 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._allreduce_delay = {
             v: self._backward_passes_per_step for _, v in sorted(named_parameters)
         }
+        # bucket gradients in reverse registration order ≈ backward firing
+        # order, so early buckets complete (and fire) while backward still
+        # runs through the front of the model
+        self._buckets = _FlatBuckets(
+            list(reversed(sorted(named_parameters))), fusion_threshold_bytes()
+        )
+        self._bucket_of = {}
+        self._pending = {}  # bucket name -> set of params still to fire
+        with torch.no_grad():
+            for i, b in enumerate(self._buckets.buckets):
+                b["name"] = f"gradbucket.{i}"
+                b["grad_flat"] = torch.zeros_like(b["flat"])
+                off = 0
+                for q in b["params"]:
+                    n = q.numel()
+                    seg = b["grad_flat"].narrow(0, off, n)
+                    q.grad = torch.as_strided(seg, q.shape, q.data.stride())
+                    off += n
+                    self._bucket_of[q] = b
         self._timeline_hook_handles = []
         self._use_timeline = False
         if _bf().size() > 1:
@@ -222,6 +245,11 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 if p.requires_grad:
                     p.register_post_accumulate_grad_hook(self._make_hook())
 
+    def _fire_bucket(self, b):
+        self._handles[b["name"]] = _bf().allreduce_nonblocking_(
+            b["grad_flat"], average=True, name=b["name"]
+        )
+
     def _make_hook(self):
         def hook(p):
             if self._allreduce_delay[p] <= 0:
@@ -230,10 +258,15 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                     self._error_encountered = True
             self._allreduce_delay[p] -= 1
             if self._allreduce_delay[p] == 0:
-                handle = _bf().allreduce_nonblocking_(
-                    p.grad, average=True, name=self._parameter_names.get(p)
+                b = self._bucket_of.get(p)
+                if b is None:
+                    return
+                pend = self._pending.setdefault(
+                    b["name"], set(id(q) for q in b["params"])
                 )
-                self._handles[p] = handle
+                pend.discard(id(p))
+                if not pend and b["name"] not in self._handles:
+                    self._fire_bucket(b)
 
         return hook
 
@@ -244,10 +277,19 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         self._use_timeline = False
 
     def synchronize(self):
+        bf = _bf()
         with torch.no_grad():
-            for p, handle in self._handles.items():
-                _bf().synchronize(handle)
+            if bf.size() > 1:
+                # flush buckets whose parameters never all fired (frozen /
+                # unused params): the collective must still run on every rank
+                for b in self._buckets.buckets:
+                    if b["name"] not in self._handles:
+                        self._fire_bucket(b)
+                for name, handle in self._handles.items():
+                    bf.synchronize(handle)
+            for p in self._allreduce_delay:
                 self._allreduce_delay[p] = self._backward_passes_per_step
+        self._pending.clear()
         self._handles.clear()
         self._synchronized = True
 
@@ -277,7 +319,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                 "optimizer.zero_grad() was called after loss.backward() but "
                 "before optimizer.step() or optimizer.synchronize()."
             )
-        return super(self.__class__, self).zero_grad(set_to_none)
+        # gradients are views into per-bucket flat buffers
+        for b in self._buckets.buckets:
+            b["grad_flat"].zero_()
+        return None
 
 
 # ---------------------------------------------------------------------------
