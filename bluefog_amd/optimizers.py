@@ -651,8 +651,115 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
         self._step_func = None
         self._timeline_hook_handles = []
         self._use_timeline = False
-        if _bf().size() > 1:
-            self._register_hooks()
+        # fused-bucket fast path, mirroring the AWC one: when a bucket's
+        # gradients have all accumulated, ONE kernel applies the SGD/Adam
+        # update to the whole bucket and ONE batched RCCL exchange ships it
+        # — O(buckets) instead of O(params) work per step, while keeping
+        # ATC's adapt-then-combine ordering and backward overlap
+        self._buckets = _FlatBuckets(
+            list(reversed(sorted(named_parameters))), fusion_threshold_bytes()
+        )
+        self._bucket_of = {}
+        self._pending = {}
+        self._fused = self._init_fused_mode()
+        # hooks apply the optimizer update itself (step() never calls the
+        # base optimizer), so they must run even at world size 1
+        self._register_hooks()
+
+    def _init_fused_mode(self):
+        if os.environ.get("BLUEFOG_FUSED_STEP", "1") in ("0", "false"):
+            return None
+        if not self._buckets.buckets:
+            return None
+        dev = self._buckets.buckets[0]["flat"].device
+        if dev.type != "cuda":
+            return None
+        from bluefog_amd.ops import hip_ext
+
+        if not hip_ext.has_extension():
+            return None
+        if len(self.param_groups) != 1:
+            return None
+        base = type(self).__mro__[1]
+        group = self.param_groups[0]
+        if base is torch.optim.SGD and group.get("dampening", 0) == 0:
+            kind = "sgd"
+        elif base is torch.optim.Adam and not group.get("amsgrad", False):
+            kind = "adam"
+        else:
+            return None
+        if self._communication_type not in (
+            CommunicationType.neighbor_allreduce,
+            CommunicationType.empty,
+        ):
+            return None
+        with torch.no_grad():
+            for b in self._buckets.buckets:
+                b["grad_flat"] = torch.zeros_like(b["flat"])
+                off = 0
+                for q in b["params"]:
+                    n = q.numel()
+                    seg = b["grad_flat"].narrow(0, off, n)
+                    q.grad = torch.as_strided(seg, q.shape, q.data.stride())
+                    off += n
+                    self._bucket_of[q] = b
+                if kind == "sgd":
+                    b["momentum_flat"] = None
+                else:
+                    b["exp_avg"] = torch.zeros(
+                        b["flat"].numel(), dtype=torch.float32, device=dev
+                    )
+                    b["exp_avg_sq"] = torch.zeros(
+                        b["flat"].numel(), dtype=torch.float32, device=dev
+                    )
+                    b["adam_step"] = 0
+        logger.debug("bluefog_amd: fused ATC %s step enabled", kind)
+        return kind
+
+    def _fused_bucket_step(self, b):
+        """Plain optimizer step over the bucket (0 neighbors) via the fused
+        kernel, then launch the bucket's neighbor exchange."""
+        from bluefog_amd import _C
+
+        group = self.param_groups[0]
+        flat, grad = b["flat"], b["grad_flat"]
+        if self._fused == "sgd":
+            momentum = group["momentum"]
+            if momentum != 0 and b["momentum_flat"] is None:
+                b["momentum_flat"] = torch.zeros_like(flat)
+            _C.weighted_combine_sgd(
+                flat, 1.0, flat, [], grad,
+                b["momentum_flat"] if momentum != 0 else flat.new_empty(0),
+                float(group["lr"]), float(momentum),
+                float(group["weight_decay"]),
+                float(group.get("dampening", 0.0)),
+                bool(group.get("nesterov", False)),
+            )
+        else:
+            b["adam_step"] += 1
+            beta1, beta2 = group["betas"]
+            _C.weighted_combine_adam(
+                flat, 1.0, flat, [], grad, b["exp_avg"], b["exp_avg_sq"],
+                float(group["lr"]), float(beta1), float(beta2),
+                float(group["eps"]), float(group["weight_decay"]),
+                int(b["adam_step"]),
+            )
+        if (
+            _bf().size() > 1
+            and self._communication_type == CommunicationType.neighbor_allreduce
+        ):
+            from bluefog_amd.ops import neighbor
+
+            self._handles[b["name"]] = (
+                "fused",
+                *neighbor.post_neighbor_exchange_raw(
+                    flat,
+                    self.self_weight,
+                    self.src_weights,
+                    self.dst_weights,
+                    self.enable_topo_check,
+                ),
+            )
 
     def _bluefog_base_step(self, closure=None):
         return super(self.__class__, self).step(closure)
@@ -688,6 +795,15 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
             self._reduce_delay[p] -= 1
             if self._reduce_delay[p] == 0:
                 with torch.no_grad():
+                    if self._fused is not None and self._step_func is None:
+                        b = self._bucket_of.get(p)
+                        pend = self._pending.setdefault(
+                            b["name"], set(id(q) for q in b["params"])
+                        )
+                        pend.discard(id(p))
+                        if not pend and b["name"] not in self._handles:
+                            self._fused_bucket_step(b)
+                        return
                     if self._step_func is not None:
                         self._step_func(p, p.grad, param_group)
                     else:
@@ -821,6 +937,8 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
     # -- communication -----------------------------------------------------
     def _launch_comm_for(self, p):
         bf = _bf()
+        if bf.size() == 1:
+            return None
         name = self._parameter_names.get(p)
         if self._communication_type == CommunicationType.allreduce:
             return bf.allreduce_nonblocking(p.data, average=True, name=name)
@@ -854,12 +972,25 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
 
     def synchronize(self):
         bf = _bf()
+        from bluefog_amd.ops import hip_ext
+
         with torch.no_grad():
-            for p, handle in self._handles.items():
-                if handle is not None:
+            for key, handle in self._handles.items():
+                if isinstance(handle, tuple) and handle and handle[0] == "fused":
+                    _, works, gathered, weights, self_w, keep = handle
+                    for w in works:
+                        w.wait()  # stream-ordered
+                    flat = next(
+                        b["flat"] for b in self._buckets.buckets if b["name"] == key
+                    )
+                    # in-place fold: flat = self_w*flat + sum w_k*recv_k
+                    hip_ext.weighted_combine(flat, flat, self_w, gathered, weights)
+                elif handle is not None:
                     output = bf.synchronize(handle)
-                    p.set_(output)
+                    key.set_(output)
+            for p in self._reduce_delay:
                 self._reduce_delay[p] = self._backward_passes_per_step
+        self._pending.clear()
         self._handles.clear()
         self._synchronized = True
 
@@ -893,6 +1024,10 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
                 "optimizer.zero_grad() was called after loss.backward() but "
                 "before optimizer.step() or optimizer.synchronize()."
             )
+        if self._fused is not None:
+            for b in self._buckets.buckets:
+                b["grad_flat"].zero_()
+            return None
         return super(self.__class__, self).zero_grad(set_to_none)
 
 
