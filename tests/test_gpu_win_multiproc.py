@@ -151,14 +151,14 @@ w_true = torch.arange(32, dtype=torch.float32).to(dev) / 32.0
 Y = (X @ w_true).unsqueeze(1) + 0.01 * torch.randn(256, 1, generator=g).to(dev)
 
 loss0 = None
-for step in range(30):
+for step in range(60):
     opt.zero_grad()
     loss = ((model(X) - Y) ** 2).mean()
     loss.backward()
     opt.step()
     if step == 0:
-        loss0 = float(loss)
-final = float(loss)
+        loss0 = float(loss.detach())
+final = float(loss.detach())
 assert final < loss0 * 0.2, (loss0, final)
 opt.unregister_window()
 print(f"WINPUT_OPT_OK rank={rank} loss {loss0:.4f}->{final:.4f}", flush=True)
