@@ -45,6 +45,11 @@ def main():
             "horovod",  # alias of gradient_allreduce, reference flag parity
         ],
     )
+    p.add_argument("--atc-style", action="store_true",
+                   help="use DistributedAdaptThenCombineOptimizer")
+    p.add_argument("--num-classes", type=int, default=1000)
+    p.add_argument("--profiler", action="store_true",
+                   help="trace one timed iteration with torch.profiler")
     p.add_argument("--disable-dynamic-topology", action="store_true")
     p.add_argument("--no-cuda", action="store_true")
     args = p.parse_args()
@@ -60,7 +65,7 @@ def main():
 
     from bluefog_amd import models
 
-    model = getattr(models, args.model)().to(device)
+    model = getattr(models, args.model)(num_classes=args.num_classes).to(device)
     if device.type == "cuda":
         model = model.to(memory_format=torch.channels_last)
 
@@ -69,6 +74,11 @@ def main():
     bf.set_topology(topo)
 
     name = args.dist_optimizer
+    awc_or_atc = (
+        bf.DistributedAdaptThenCombineOptimizer
+        if args.atc_style
+        else bf.DistributedAdaptWithCombineOptimizer
+    )
     if name in ("gradient_allreduce", "horovod"):
         optimizer = bf.DistributedGradientAllreduceOptimizer(base_opt, model=model)
     elif name == "win_put":
@@ -77,7 +87,7 @@ def main():
         optimizer = bf.DistributedPushSumOptimizer(base_opt, model=model)
     elif name == "hierarchical_neighbor_allreduce":
         bf.set_machine_topology(bf.ExponentialTwoGraph(bf.machine_size()))
-        optimizer = bf.DistributedAdaptWithCombineOptimizer(
+        optimizer = awc_or_atc(
             base_opt, model=model,
             communication_type=bf.CommunicationType.hierarchical_neighbor_allreduce,
         )
@@ -87,9 +97,7 @@ def main():
             if name == "allreduce"
             else bf.CommunicationType.neighbor_allreduce
         )
-        optimizer = bf.DistributedAdaptWithCombineOptimizer(
-            base_opt, model=model, communication_type=ct
-        )
+        optimizer = awc_or_atc(base_opt, model=model, communication_type=ct)
 
     dyn_gen = None
     if (
@@ -121,6 +129,25 @@ def main():
 
     for _ in range(args.num_warmup_batches):
         benchmark_step()
+
+    if args.profiler:
+        # reference --profiler used torch.autograd.profiler around one
+        # iteration; torch.profiler is its successor
+        from torch.profiler import ProfilerActivity, profile
+
+        acts = [ProfilerActivity.CPU]
+        if device.type == "cuda":
+            acts.append(ProfilerActivity.CUDA)
+        with profile(activities=acts) as prof:
+            for _ in range(args.num_batches_per_iter):
+                benchmark_step()
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+        if bf.rank() == 0:
+            print(prof.key_averages().table(sort_by="self_cuda_time_total"
+                                            if device.type == "cuda"
+                                            else "self_cpu_time_total",
+                                            row_limit=25))
 
     img_secs = []
     for _ in range(args.num_iters):
