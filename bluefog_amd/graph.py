@@ -12,7 +12,7 @@ A real ``networkx.DiGraph`` is accepted anywhere a :class:`DiGraph` is, via
 :func:`as_digraph` duck-typing conversion.
 """
 
-from typing import Iterable, List, Optional, Tuple
+from typing import Iterable, List, Optional
 
 import numpy as np
 

@@ -4,7 +4,7 @@
 self-contained; uses torch's fused scaled_dot_product_attention, which on
 ROCm lowers to the MIOpen/CK flash-attention path."""
 
-import math
+
 
 import torch
 import torch.nn as nn

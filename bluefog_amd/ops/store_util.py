@@ -17,12 +17,12 @@ Control traffic is a few hundred bytes per acquire/release — entirely off
 the xGMI data plane.
 """
 
-import os
+
 import time
 from datetime import timedelta
 from typing import List, Optional
 
-import torch.distributed as dist
+
 
 
 class ControlStore:

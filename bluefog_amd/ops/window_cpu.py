@@ -18,7 +18,7 @@ import socket
 import socketserver
 import struct
 import threading
-from typing import Optional
+
 
 import torch
 

@@ -19,7 +19,7 @@ import os
 import warnings
 from contextlib import contextmanager
 from enum import Enum
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 
