@@ -13,6 +13,7 @@ Protocol: length-prefixed pickled dicts, one request per connection.
 """
 
 import io
+import os
 import pickle
 import socket
 import socketserver
@@ -75,7 +76,9 @@ class WindowServer:
 
         self._server = Server(("0.0.0.0", 0), Handler)
         self.port = self._server.server_address[1]
-        self.host = socket.gethostname()
+        # single-node by design; container hostnames often do not resolve
+        # (use the loopback, matching the rendezvous contract)
+        self.host = os.environ.get("BLUEFOG_WIN_SERVER_HOST", "127.0.0.1")
         self._thread = threading.Thread(target=self._server.serve_forever, daemon=True)
         self._thread.start()
 
