@@ -1,0 +1,57 @@
+# Copyright 2026. Licensed under the Apache License, Version 2.0.
+"""Guard the driver's bench.py contract: one JSON line from rank 0 with the
+required keys and the BASELINE metric/config names."""
+
+import json
+import os
+import subprocess
+import sys
+
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract():
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(_ROOT, "bench.py"),
+         "--model", "resnet18", "--batch-size", "2", "--steps", "2",
+         "--warmup", "1", "--device", "cpu"],
+        cwd=_ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in rec, key
+    assert rec["n_gpus"] == 1
+    assert rec["steps"] == 2
+    assert rec["data"] == "synthetic"
+    assert rec["scaling"] == "weak"
+    assert rec["value"] > 0
+    assert set(rec["config"]) >= {"model", "global_batch", "seq_len", "parallelism"}
+
+
+def test_bench_default_metric_is_baseline_config():
+    """Default invocation must measure the BASELINE.json headline config."""
+    import argparse
+
+    sys.path.insert(0, _ROOT)
+    import bench
+
+    args = bench.parse_args.__wrapped__() if hasattr(bench.parse_args, "__wrapped__") else None
+    # parse with no CLI args
+    old = sys.argv
+    sys.argv = ["bench.py"]
+    try:
+        args = bench.parse_args()
+    finally:
+        sys.argv = old
+    assert args.model == "resnet50"
+    assert args.batch_size == 64
+    assert args.dist_optimizer == "neighbor_allreduce"
+    assert args.dtype == "fp32"
