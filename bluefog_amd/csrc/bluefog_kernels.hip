@@ -389,6 +389,148 @@ __global__ __launch_bounds__(kThreads) void relu_bwd_mask_k(
     gin[t] = AccOf<T>::to(out[t]) > static_cast<Acc>(0) ? g[t] : AccOf<T>::from(0.f);
 }
 
+// ---------------------------------------------------------------------------
+// Fused training BatchNorm2d + ReLU (NHWC or NCHW dense layouts).
+//
+// Replaces the MIOpen BN kernel chain + separate ReLU clamp/threshold
+// kernels in the ResNet hot path with minimal-traffic passes:
+//   fwd:  stats (1 read of x)  ->  finalize (C)  ->  normalize+ReLU (1R 1W)
+//   bwd:  reduce (x,dy: 2R)    ->  finalize (C)  ->  dx (x,dy: 2R 1W)
+// The ReLU mask is recomputed from x (gamma*xhat+beta > 0), so neither the
+// forward output nor a mask tensor is re-read in backward.
+// Per-channel accumulation: one LDS array per workgroup, one global
+// atomicAdd per (channel, workgroup) — C <= kMaxBnChannels floats of LDS.
+// ---------------------------------------------------------------------------
+
+constexpr int kMaxBnChannels = 4096;  // 2 fp32 arrays * 4096 = 32 KB LDS
+
+template <typename T, bool NHWC>
+__device__ __forceinline__ int bn_channel_of(long i, int C, long HW) {
+  return NHWC ? static_cast<int>(i % C) : static_cast<int>((i / HW) % C);
+}
+
+template <typename T, bool NHWC>
+__global__ __launch_bounds__(kThreads) void bn_stats_k(
+    const T* __restrict__ x, float* __restrict__ ws /* [2C]: sum, sumsq */,
+    int C, long HW, long numel) {
+  extern __shared__ float lds[];  // [2C]
+  float* lsum = lds;
+  float* lsq = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    const float v = AccOf<T>::to(x[i]);
+    const int c = bn_channel_of<T, NHWC>(i, C, HW);
+    atomicAdd(&lsum[c], v);
+    atomicAdd(&lsq[c], v * v);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lsum[c] != 0.f) atomicAdd(&ws[c], lsum[c]);
+    if (lsq[c] != 0.f) atomicAdd(&ws[C + c], lsq[c]);
+  }
+}
+
+__global__ __launch_bounds__(256) void bn_fwd_finalize_k(
+    const float* __restrict__ ws, const float* __restrict__ running_mean_in,
+    const float* __restrict__ running_var_in, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd, float momentum, float eps, float count,
+    int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mean = ws[c] / count;
+  float var = ws[C + c] / count - mean * mean;
+  if (var < 0.f) var = 0.f;
+  save_mean[c] = mean;
+  save_invstd[c] = rsqrtf(var + eps);
+  const float unbiased = count > 1.f ? var * count / (count - 1.f) : var;
+  running_mean[c] = (1.f - momentum) * running_mean_in[c] + momentum * mean;
+  running_var[c] = (1.f - momentum) * running_var_in[c] + momentum * unbiased;
+}
+
+template <typename T, bool NHWC>
+__global__ __launch_bounds__(kThreads) void bn_fwd_norm_relu_k(
+    T* __restrict__ y, const T* __restrict__ x,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ mean, const float* __restrict__ invstd, int C,
+    long HW, long numel) {
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    const int c = bn_channel_of<T, NHWC>(i, C, HW);
+    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
+    float v = gamma[c] * xhat + beta[c];
+    y[i] = AccOf<T>::from(v > 0.f ? v : 0.f);
+  }
+}
+
+template <typename T, bool NHWC>
+__global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ ws /* [2C]: sum_g, sum_g_xhat */, int C, long HW,
+    long numel) {
+  extern __shared__ float lds[];
+  float* lg = lds;
+  float* lgx = lds + C;
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    const int c = bn_channel_of<T, NHWC>(i, C, HW);
+    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
+    const float pre = gamma[c] * xhat + beta[c];
+    if (pre > 0.f) {
+      const float g = AccOf<T>::to(dy[i]);
+      atomicAdd(&lg[c], g);
+      atomicAdd(&lgx[c], g * xhat);
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lg[c] != 0.f) atomicAdd(&ws[c], lg[c]);
+    if (lgx[c] != 0.f) atomicAdd(&ws[C + c], lgx[c]);
+  }
+}
+
+__global__ __launch_bounds__(256) void bn_bwd_finalize_k(
+    const float* __restrict__ ws, const float* __restrict__ gamma,
+    const float* __restrict__ invstd, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, float* __restrict__ coef /* [3C]: a,b,c */,
+    float count, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float sum_g = ws[c];
+  const float sum_gx = ws[C + c];
+  dbeta[c] = sum_g;
+  dgamma[c] = sum_gx;
+  coef[c] = gamma[c] * invstd[c];           // a
+  coef[C + c] = sum_g / count;              // b (mean of g)
+  coef[2 * C + c] = sum_gx / count;         // c (mean of g*xhat)
+}
+
+template <typename T, bool NHWC>
+__global__ __launch_bounds__(kThreads) void bn_bwd_dx_k(
+    T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ coef, int C, long HW, long numel) {
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < numel; i += stride) {
+    const int c = bn_channel_of<T, NHWC>(i, C, HW);
+    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
+    const float pre = gamma[c] * xhat + beta[c];
+    const float g = pre > 0.f ? AccOf<T>::to(dy[i]) : 0.f;
+    dx[i] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
+  }
+}
+
 enum BfDtype : int { kF32 = 0, kF64 = 1, kF16 = 2, kBF16 = 3 };
 
 template <template <typename> class Fn, typename... Args>
@@ -629,6 +771,89 @@ struct ReluBwdMaskLauncher {
   }
 };
 
+inline int bn_grid(long numel) {
+  long blocks = (numel + kThreads * 64 - 1) / (kThreads * 64);
+  if (blocks > 1024) blocks = 1024;
+  if (blocks < 8) blocks = 8;
+  return static_cast<int>(blocks);
+}
+
+template <typename T>
+struct BnFwdLauncher {
+  static hipError_t run(void* y, const void* x, const float* gamma,
+                        const float* beta, const float* run_mean_in,
+                        const float* run_var_in, float* run_mean,
+                        float* run_var, float* save_mean, float* save_invstd,
+                        float* ws, float momentum, float eps, int C, long HW,
+                        long numel, bool nhwc, hipStream_t stream) {
+    const int sgrid = bn_grid(numel);
+    const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    if (nhwc)
+      hipLaunchKernelGGL((bn_stats_k<T, true>), dim3(sgrid), dim3(kThreads), lds,
+                         stream, static_cast<const T*>(x), ws, C, HW, numel);
+    else
+      hipLaunchKernelGGL((bn_stats_k<T, false>), dim3(sgrid), dim3(kThreads), lds,
+                         stream, static_cast<const T*>(x), ws, C, HW, numel);
+    BF_CHECK_HIP(hipGetLastError());
+    const float count = static_cast<float>(numel / C);
+    hipLaunchKernelGGL(bn_fwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
+                       stream, ws, run_mean_in, run_var_in, run_mean, run_var,
+                       save_mean, save_invstd, momentum, eps, count, C);
+    BF_CHECK_HIP(hipGetLastError());
+    const int ngrid = grid_for(numel);
+    if (nhwc)
+      hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, true>), dim3(ngrid), dim3(kThreads),
+                         0, stream, static_cast<T*>(y), static_cast<const T*>(x),
+                         gamma, beta, save_mean, save_invstd, C, HW, numel);
+    else
+      hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, false>), dim3(ngrid), dim3(kThreads),
+                         0, stream, static_cast<T*>(y), static_cast<const T*>(x),
+                         gamma, beta, save_mean, save_invstd, C, HW, numel);
+    return hipGetLastError();
+  }
+};
+
+template <typename T>
+struct BnBwdLauncher {
+  static hipError_t run(void* dx, const void* x, const void* dy,
+                        const float* gamma, const float* beta,
+                        const float* save_mean, const float* save_invstd,
+                        float* ws, float* dgamma, float* dbeta, float* coef,
+                        int C, long HW, long numel, bool nhwc,
+                        hipStream_t stream) {
+    const int sgrid = bn_grid(numel);
+    const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    if (nhwc)
+      hipLaunchKernelGGL((bn_bwd_reduce_k<T, true>), dim3(sgrid), dim3(kThreads),
+                         lds, stream, static_cast<const T*>(x),
+                         static_cast<const T*>(dy), gamma, beta, save_mean,
+                         save_invstd, ws, C, HW, numel);
+    else
+      hipLaunchKernelGGL((bn_bwd_reduce_k<T, false>), dim3(sgrid), dim3(kThreads),
+                         lds, stream, static_cast<const T*>(x),
+                         static_cast<const T*>(dy), gamma, beta, save_mean,
+                         save_invstd, ws, C, HW, numel);
+    BF_CHECK_HIP(hipGetLastError());
+    const float count = static_cast<float>(numel / C);
+    hipLaunchKernelGGL(bn_bwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
+                       stream, ws, gamma, save_invstd, dgamma, dbeta, coef,
+                       count, C);
+    BF_CHECK_HIP(hipGetLastError());
+    const int ngrid = grid_for(numel);
+    if (nhwc)
+      hipLaunchKernelGGL((bn_bwd_dx_k<T, true>), dim3(ngrid), dim3(kThreads), 0,
+                         stream, static_cast<T*>(dx), static_cast<const T*>(x),
+                         static_cast<const T*>(dy), gamma, beta, save_mean,
+                         save_invstd, coef, C, HW, numel);
+    else
+      hipLaunchKernelGGL((bn_bwd_dx_k<T, false>), dim3(ngrid), dim3(kThreads), 0,
+                         stream, static_cast<T*>(dx), static_cast<const T*>(x),
+                         static_cast<const T*>(dy), gamma, beta, save_mean,
+                         save_invstd, coef, C, HW, numel);
+    return hipGetLastError();
+  }
+};
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -685,6 +910,30 @@ hipError_t bf_add_relu_fwd(void* out, const void* a, const void* b,
 hipError_t bf_relu_bwd_mask(void* gin, const void* g, const void* out,
                             long numel, int dtype, hipStream_t stream) {
   return dispatch_dtype<ReluBwdMaskLauncher>(dtype, gin, g, out, numel, stream);
+}
+
+hipError_t bf_bn_relu_fwd(void* y, const void* x, const float* gamma,
+                          const float* beta, const float* run_mean_in,
+                          const float* run_var_in, float* run_mean,
+                          float* run_var, float* save_mean, float* save_invstd,
+                          float* ws, float momentum, float eps, int C, long HW,
+                          long numel, bool nhwc, int dtype,
+                          hipStream_t stream) {
+  return dispatch_dtype<BnFwdLauncher>(dtype, y, x, gamma, beta, run_mean_in,
+                                       run_var_in, run_mean, run_var, save_mean,
+                                       save_invstd, ws, momentum, eps, C, HW,
+                                       numel, nhwc, stream);
+}
+
+hipError_t bf_bn_relu_bwd(void* dx, const void* x, const void* dy,
+                          const float* gamma, const float* beta,
+                          const float* save_mean, const float* save_invstd,
+                          float* ws, float* dgamma, float* dbeta, float* coef,
+                          int C, long HW, long numel, bool nhwc, int dtype,
+                          hipStream_t stream) {
+  return dispatch_dtype<BnBwdLauncher>(dtype, dx, x, dy, gamma, beta, save_mean,
+                                       save_invstd, ws, dgamma, dbeta, coef, C,
+                                       HW, numel, nhwc, stream);
 }
 
 }  // extern "C"
