@@ -409,7 +409,7 @@ __device__ __forceinline__ int bn_channel_of(long i, int C, long HW) {
   return NHWC ? static_cast<int>(i % C) : static_cast<int>((i / HW) % C);
 }
 
-template <typename T, bool NHWC>
+template <typename T, bool NHWC, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_stats_k(
     const T* __restrict__ x, float* __restrict__ ws /* [2C]: sum, sumsq */,
     int C, long HW, long numel) {
@@ -418,13 +418,26 @@ __global__ __launch_bounds__(kThreads) void bn_stats_k(
   float* lsq = lds + C;
   for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
   __syncthreads();
+  const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < numel; i += stride) {
-    const float v = AccOf<T>::to(x[i]);
-    const int c = bn_channel_of<T, NHWC>(i, C, HW);
-    atomicAdd(&lsum[c], v);
-    atomicAdd(&lsq[c], v * v);
+       i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const float val = AccOf<T>::to(xv.v[v]);
+      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
+      atomicAdd(&lsum[c], val);
+      atomicAdd(&lsq[c], val * val);
+    }
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    const float val = AccOf<T>::to(x[t]);
+    const int c = bn_channel_of<T, NHWC>(t, C, HW);
+    atomicAdd(&lsum[c], val);
+    atomicAdd(&lsq[c], val * val);
   }
   __syncthreads();
   for (int c = threadIdx.x; c < C; c += blockDim.x) {
@@ -434,7 +447,7 @@ __global__ __launch_bounds__(kThreads) void bn_stats_k(
 }
 
 __global__ __launch_bounds__(256) void bn_fwd_finalize_k(
-    const float* __restrict__ ws, const float* __restrict__ running_mean_in,
+    float* __restrict__ ws, const float* __restrict__ running_mean_in,
     const float* __restrict__ running_var_in, float* __restrict__ running_mean,
     float* __restrict__ running_var, float* __restrict__ save_mean,
     float* __restrict__ save_invstd, float momentum, float eps, float count,
@@ -444,6 +457,8 @@ __global__ __launch_bounds__(256) void bn_fwd_finalize_k(
   const float mean = ws[c] / count;
   float var = ws[C + c] / count - mean * mean;
   if (var < 0.f) var = 0.f;
+  ws[c] = 0.f;      // reset for the next launch (module reuses the buffer)
+  ws[C + c] = 0.f;
   save_mean[c] = mean;
   save_invstd[c] = rsqrtf(var + eps);
   const float unbiased = count > 1.f ? var * count / (count - 1.f) : var;
@@ -451,23 +466,37 @@ __global__ __launch_bounds__(256) void bn_fwd_finalize_k(
   running_var[c] = (1.f - momentum) * running_var_in[c] + momentum * unbiased;
 }
 
-template <typename T, bool NHWC>
+template <typename T, bool NHWC, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_fwd_norm_relu_k(
     T* __restrict__ y, const T* __restrict__ x,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ mean, const float* __restrict__ invstd, int C,
     long HW, long numel) {
+  const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < numel; i += stride) {
-    const int c = bn_channel_of<T, NHWC>(i, C, HW);
-    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
+       i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
+      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
+      const float val = gamma[c] * xhat + beta[c];
+      xv.v[v] = AccOf<T>::from(val > 0.f ? val : 0.f);
+    }
+    *reinterpret_cast<Pack<T, VEC>*>(y + base) = xv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    const int c = bn_channel_of<T, NHWC>(t, C, HW);
+    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
     float v = gamma[c] * xhat + beta[c];
-    y[i] = AccOf<T>::from(v > 0.f ? v : 0.f);
+    y[t] = AccOf<T>::from(v > 0.f ? v : 0.f);
   }
 }
 
-template <typename T, bool NHWC>
+template <typename T, bool NHWC, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
     const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -479,14 +508,30 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
   float* lgx = lds + C;
   for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
   __syncthreads();
+  const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < numel; i += stride) {
-    const int c = bn_channel_of<T, NHWC>(i, C, HW);
-    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
-    const float pre = gamma[c] * xhat + beta[c];
-    if (pre > 0.f) {
-      const float g = AccOf<T>::to(dy[i]);
+       i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
+    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
+      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
+      if (gamma[c] * xhat + beta[c] > 0.f) {
+        const float g = AccOf<T>::to(gv.v[v]);
+        atomicAdd(&lg[c], g);
+        atomicAdd(&lgx[c], g * xhat);
+      }
+    }
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    const int c = bn_channel_of<T, NHWC>(t, C, HW);
+    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
+    if (gamma[c] * xhat + beta[c] > 0.f) {
+      const float g = AccOf<T>::to(dy[t]);
       atomicAdd(&lg[c], g);
       atomicAdd(&lgx[c], g * xhat);
     }
@@ -499,7 +544,7 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
 }
 
 __global__ __launch_bounds__(256) void bn_bwd_finalize_k(
-    const float* __restrict__ ws, const float* __restrict__ gamma,
+    float* __restrict__ ws, const float* __restrict__ gamma,
     const float* __restrict__ invstd, float* __restrict__ dgamma,
     float* __restrict__ dbeta, float* __restrict__ coef /* [3C]: a,b,c */,
     float count, int C) {
@@ -507,6 +552,8 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_k(
   if (c >= C) return;
   const float sum_g = ws[c];
   const float sum_gx = ws[C + c];
+  ws[c] = 0.f;
+  ws[C + c] = 0.f;
   dbeta[c] = sum_g;
   dgamma[c] = sum_gx;
   coef[c] = gamma[c] * invstd[c];           // a
@@ -514,20 +561,36 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_k(
   coef[2 * C + c] = sum_gx / count;         // c (mean of g*xhat)
 }
 
-template <typename T, bool NHWC>
+template <typename T, bool NHWC, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_bwd_dx_k(
     T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ coef, int C, long HW, long numel) {
+  const long nvec = numel / VEC;
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < numel; i += stride) {
-    const int c = bn_channel_of<T, NHWC>(i, C, HW);
-    const float xhat = (AccOf<T>::to(x[i]) - mean[c]) * invstd[c];
+       i < nvec; i += stride) {
+    const long base = i * VEC;
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
+    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
+      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
+      const float g =
+          gamma[c] * xhat + beta[c] > 0.f ? AccOf<T>::to(gv.v[v]) : 0.f;
+      xv.v[v] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
+    }
+    *reinterpret_cast<Pack<T, VEC>*>(dx + base) = xv;
+  }
+  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
+       t < numel; t += stride) {
+    const int c = bn_channel_of<T, NHWC>(t, C, HW);
+    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
     const float pre = gamma[c] * xhat + beta[c];
-    const float g = pre > 0.f ? AccOf<T>::to(dy[i]) : 0.f;
-    dx[i] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
+    const float g = pre > 0.f ? AccOf<T>::to(dy[t]) : 0.f;
+    dx[t] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
   }
 }
 
@@ -786,29 +849,32 @@ struct BnFwdLauncher {
                         float* run_var, float* save_mean, float* save_invstd,
                         float* ws, float momentum, float eps, int C, long HW,
                         long numel, bool nhwc, hipStream_t stream) {
-    const int sgrid = bn_grid(numel);
+    constexpr int V = vec_width<T>();
+    const bool vec = vec_ok<T>(x, numel);
+    const int sgrid = bn_grid(numel / (vec ? V : 1));
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
-    if (nhwc)
-      hipLaunchKernelGGL((bn_stats_k<T, true>), dim3(sgrid), dim3(kThreads), lds,
-                         stream, static_cast<const T*>(x), ws, C, HW, numel);
-    else
-      hipLaunchKernelGGL((bn_stats_k<T, false>), dim3(sgrid), dim3(kThreads), lds,
-                         stream, static_cast<const T*>(x), ws, C, HW, numel);
+#define BF_BN_STATS(NH, VV)                                                    \
+  hipLaunchKernelGGL((bn_stats_k<T, NH, VV>), dim3(sgrid), dim3(kThreads),     \
+                     lds, stream, static_cast<const T*>(x), ws, C, HW, numel)
+    if (nhwc) { if (vec) BF_BN_STATS(true, V); else BF_BN_STATS(true, 1); }
+    else      { if (vec) BF_BN_STATS(false, V); else BF_BN_STATS(false, 1); }
+#undef BF_BN_STATS
     BF_CHECK_HIP(hipGetLastError());
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_fwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
                        stream, ws, run_mean_in, run_var_in, run_mean, run_var,
                        save_mean, save_invstd, momentum, eps, count, C);
     BF_CHECK_HIP(hipGetLastError());
-    const int ngrid = grid_for(numel);
-    if (nhwc)
-      hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, true>), dim3(ngrid), dim3(kThreads),
-                         0, stream, static_cast<T*>(y), static_cast<const T*>(x),
-                         gamma, beta, save_mean, save_invstd, C, HW, numel);
-    else
-      hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, false>), dim3(ngrid), dim3(kThreads),
-                         0, stream, static_cast<T*>(y), static_cast<const T*>(x),
-                         gamma, beta, save_mean, save_invstd, C, HW, numel);
+    const bool nvecok = vec && vec_ok<T>(y, numel);
+    const int ngrid = grid_for((numel + (nvecok ? V : 1) - 1) / (nvecok ? V : 1));
+#define BF_BN_NORM(NH, VV)                                                     \
+  hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, NH, VV>), dim3(ngrid),             \
+                     dim3(kThreads), 0, stream, static_cast<T*>(y),            \
+                     static_cast<const T*>(x), gamma, beta, save_mean,         \
+                     save_invstd, C, HW, numel)
+    if (nhwc) { if (nvecok) BF_BN_NORM(true, V); else BF_BN_NORM(true, 1); }
+    else      { if (nvecok) BF_BN_NORM(false, V); else BF_BN_NORM(false, 1); }
+#undef BF_BN_NORM
     return hipGetLastError();
   }
 };
@@ -821,35 +887,34 @@ struct BnBwdLauncher {
                         float* ws, float* dgamma, float* dbeta, float* coef,
                         int C, long HW, long numel, bool nhwc,
                         hipStream_t stream) {
-    const int sgrid = bn_grid(numel);
+    constexpr int V = vec_width<T>();
+    const bool vec = vec_ok<T>(x, numel) && vec_ok<T>(dy, numel);
+    const int sgrid = bn_grid(numel / (vec ? V : 1));
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
-    if (nhwc)
-      hipLaunchKernelGGL((bn_bwd_reduce_k<T, true>), dim3(sgrid), dim3(kThreads),
-                         lds, stream, static_cast<const T*>(x),
-                         static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, ws, C, HW, numel);
-    else
-      hipLaunchKernelGGL((bn_bwd_reduce_k<T, false>), dim3(sgrid), dim3(kThreads),
-                         lds, stream, static_cast<const T*>(x),
-                         static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, ws, C, HW, numel);
+#define BF_BN_RED(NH, VV)                                                      \
+  hipLaunchKernelGGL((bn_bwd_reduce_k<T, NH, VV>), dim3(sgrid),                \
+                     dim3(kThreads), lds, stream, static_cast<const T*>(x),    \
+                     static_cast<const T*>(dy), gamma, beta, save_mean,        \
+                     save_invstd, ws, C, HW, numel)
+    if (nhwc) { if (vec) BF_BN_RED(true, V); else BF_BN_RED(true, 1); }
+    else      { if (vec) BF_BN_RED(false, V); else BF_BN_RED(false, 1); }
+#undef BF_BN_RED
     BF_CHECK_HIP(hipGetLastError());
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_bwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
                        stream, ws, gamma, save_invstd, dgamma, dbeta, coef,
                        count, C);
     BF_CHECK_HIP(hipGetLastError());
-    const int ngrid = grid_for(numel);
-    if (nhwc)
-      hipLaunchKernelGGL((bn_bwd_dx_k<T, true>), dim3(ngrid), dim3(kThreads), 0,
-                         stream, static_cast<T*>(dx), static_cast<const T*>(x),
-                         static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, coef, C, HW, numel);
-    else
-      hipLaunchKernelGGL((bn_bwd_dx_k<T, false>), dim3(ngrid), dim3(kThreads), 0,
-                         stream, static_cast<T*>(dx), static_cast<const T*>(x),
-                         static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, coef, C, HW, numel);
+    const bool dvec = vec && vec_ok<T>(dx, numel);
+    const int ngrid = grid_for((numel + (dvec ? V : 1) - 1) / (dvec ? V : 1));
+#define BF_BN_DX(NH, VV)                                                       \
+  hipLaunchKernelGGL((bn_bwd_dx_k<T, NH, VV>), dim3(ngrid), dim3(kThreads), 0, \
+                     stream, static_cast<T*>(dx), static_cast<const T*>(x),    \
+                     static_cast<const T*>(dy), gamma, beta, save_mean,        \
+                     save_invstd, coef, C, HW, numel)
+    if (nhwc) { if (dvec) BF_BN_DX(true, V); else BF_BN_DX(true, 1); }
+    else      { if (dvec) BF_BN_DX(false, V); else BF_BN_DX(false, 1); }
+#undef BF_BN_DX
     return hipGetLastError();
   }
 };

@@ -93,7 +93,7 @@ def _bn_fusable(x: torch.Tensor, weight) -> bool:
 
 class _BNReLU(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps):
+    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps, ws):
         from bluefog_amd import _C
 
         C = x.size(1)
@@ -101,12 +101,14 @@ class _BNReLU(torch.autograd.Function):
         f32 = dict(device=x.device, dtype=torch.float32)
         save_mean = torch.empty(C, **f32)
         save_invstd = torch.empty(C, **f32)
-        ws = torch.zeros(2 * C, **f32)
+        # ws is the module's persistent [2C] fp32 workspace; the finalize
+        # kernels re-zero it after reading, so it is always clean here
         _C.bn_relu_fwd(
             y, x, weight, bias, running_mean, running_var,
             save_mean, save_invstd, ws, float(momentum), float(eps),
         )
         ctx.save_for_backward(x, weight, bias, save_mean, save_invstd)
+        ctx.ws = ws
         return y
 
     @staticmethod
@@ -125,13 +127,12 @@ class _BNReLU(torch.autograd.Function):
         dx = torch.empty_like(x)
         dgamma = torch.empty(C, **f32)
         dbeta = torch.empty(C, **f32)
-        ws = torch.zeros(2 * C, **f32)
         coef = torch.empty(3 * C, **f32)
         _C.bn_relu_bwd(
             dx, x, grad_out, weight, bias, save_mean, save_invstd,
-            ws, dgamma, dbeta, coef,
+            ctx.ws, dgamma, dbeta, coef,
         )
-        return dx, dgamma, dbeta, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 class FusedBNReLU2d(torch.nn.BatchNorm2d):
@@ -158,9 +159,15 @@ class FusedBNReLU2d(torch.nn.BatchNorm2d):
                 momentum = 1.0 / float(self.num_batches_tracked)
             else:
                 momentum = self.momentum
+            ws = getattr(self, "_bf_ws", None)
+            if ws is None or ws.device != x.device:
+                ws = torch.zeros(
+                    2 * self.num_features, device=x.device, dtype=torch.float32
+                )
+                self._bf_ws = ws
             return _BNReLU.apply(
                 x, self.weight, self.bias, self.running_mean, self.running_var,
-                momentum, self.eps,
+                momentum, self.eps, ws,
             )
         # eager fallback: in-place ReLU on the BN output (the pre-fusion
         # fast path — BN output has no other consumer)
