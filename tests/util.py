@@ -52,7 +52,36 @@ def run_dist(
     timeout: float = 180.0,
 ) -> None:
     """Run ``fn()`` (a module-level function) in ``world_size`` fresh
-    processes with a gloo/RCCL rendezvous on 127.0.0.1."""
+    processes with a gloo/RCCL rendezvous on 127.0.0.1. Rendezvous-level
+    failures (a just-freed port grabbed by another process before the
+    store binds it) get ONE retry on a fresh port; genuine test failures
+    do not."""
+    try:
+        return _run_dist_once(fn, world_size, args, env, timeout)
+    except (TimeoutError, AssertionError) as e:
+        msg = str(e)
+        transient = any(
+            pat in msg
+            for pat in (
+                "Address already in use",
+                "EADDRINUSE",
+                "Connection refused",
+                "Connection reset",
+                "timed out",
+            )
+        )
+        if not transient:
+            raise
+        return _run_dist_once(fn, world_size, args, env, timeout)
+
+
+def _run_dist_once(
+    fn: Callable,
+    world_size: int,
+    args: tuple,
+    env: Optional[Dict[str, str]],
+    timeout: float,
+) -> None:
     ctx = mp.get_context("spawn")
     errq = ctx.Queue()
     port = free_port()
@@ -77,7 +106,7 @@ def run_dist(
                 p.terminate()
             raise TimeoutError(
                 f"run_dist({fn.__name__}, world_size={world_size}) timed out"
-            )
+            )  # noqa: TRY003 — retried once by run_dist for transient causes
         done += 1
         if err is not None:
             failures.append(f"--- rank {rank} ---\n{err}")
