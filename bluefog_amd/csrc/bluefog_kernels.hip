@@ -29,6 +29,7 @@
 #include <hip/hip_bf16.h>
 
 #include <cstdint>
+#include <numeric>
 
 #define BF_CHECK_HIP(cmd)                                                     \
   do {                                                                        \
@@ -407,6 +408,81 @@ constexpr int kMaxBnChannels = 4096;  // 2 fp32 arrays * 4096 = 32 KB LDS
 template <typename T, bool NHWC>
 __device__ __forceinline__ int bn_channel_of(long i, int C, long HW) {
   return NHWC ? static_cast<int>(i % C) : static_cast<int>((i / HW) % C);
+}
+
+// NHWC fast path: the launcher sizes the grid so that the grid stride is a
+// multiple of C — each lane then touches a FIXED set of VEC channels for
+// its whole loop and accumulates in registers; only 2*VEC atomics per
+// thread at the end (vs 2 per element through LDS).
+template <typename T, int VEC>
+__global__ __launch_bounds__(kThreads) void bn_stats_nhwc_reg_k(
+    const T* __restrict__ x, float* __restrict__ ws /* [2C] */, int C,
+    long numel) {
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;  // *VEC elems
+  const long nvec = numel / VEC;
+  float sum[VEC], sq[VEC];
+#pragma unroll
+  for (int v = 0; v < VEC; ++v) sum[v] = sq[v] = 0.f;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int c0 = static_cast<int>((i * VEC) % C);  // fixed: stride*VEC % C == 0
+  for (; i < nvec; i += stride) {
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + i * VEC);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const float val = AccOf<T>::to(xv.v[v]);
+      sum[v] += val;
+      sq[v] += val * val;
+    }
+  }
+#pragma unroll
+  for (int v = 0; v < VEC; ++v) {
+    const int c = (c0 + v) % C;
+    if (sum[v] != 0.f) atomicAdd(&ws[c], sum[v]);
+    if (sq[v] != 0.f) atomicAdd(&ws[C + c], sq[v]);
+  }
+}
+
+template <typename T, int VEC>
+__global__ __launch_bounds__(kThreads) void bn_bwd_reduce_nhwc_reg_k(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ ws /* [2C] */, int C, long numel) {
+  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
+  const long nvec = numel / VEC;
+  float sg[VEC], sgx[VEC];
+#pragma unroll
+  for (int v = 0; v < VEC; ++v) sg[v] = sgx[v] = 0.f;
+  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  const int c0 = static_cast<int>((i * VEC) % C);
+  float mv[VEC], iv[VEC], gm[VEC], bt[VEC];
+#pragma unroll
+  for (int v = 0; v < VEC; ++v) {
+    const int c = (c0 + v) % C;
+    mv[v] = mean[c];
+    iv[v] = invstd[c];
+    gm[v] = gamma[c];
+    bt[v] = beta[c];
+  }
+  for (; i < nvec; i += stride) {
+    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + i * VEC);
+    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + i * VEC);
+#pragma unroll
+    for (int v = 0; v < VEC; ++v) {
+      const float xhat = (AccOf<T>::to(xv.v[v]) - mv[v]) * iv[v];
+      if (gm[v] * xhat + bt[v] > 0.f) {
+        const float g = AccOf<T>::to(gv.v[v]);
+        sg[v] += g;
+        sgx[v] += g * xhat;
+      }
+    }
+  }
+#pragma unroll
+  for (int v = 0; v < VEC; ++v) {
+    const int c = (c0 + v) % C;
+    if (sg[v] != 0.f) atomicAdd(&ws[c], sg[v]);
+    if (sgx[v] != 0.f) atomicAdd(&ws[C + c], sgx[v]);
+  }
 }
 
 template <typename T, bool NHWC, int VEC>
@@ -851,14 +927,30 @@ struct BnFwdLauncher {
                         long numel, bool nhwc, hipStream_t stream) {
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(x, numel);
-    const int sgrid = bn_grid(numel / (vec ? V : 1));
+    int sgrid = bn_grid(numel / (vec ? V : 1));
+    // register path: fixed channel per lane requires (grid*threads*V) % C == 0
+    const bool reg = nhwc && vec && (C % V == 0);
+    if (reg) {
+      const int chunks = (C + kThreads * V - 1) / (kThreads * V);
+      const int align = chunks > 1 ? chunks : 1;  // C > lane coverage
+      sgrid = ((sgrid + align - 1) / align) * align;
+      if ((static_cast<long>(sgrid) * kThreads * V) % C != 0)
+        sgrid *= C / static_cast<int>(std::gcd(static_cast<long>(C),
+                                               static_cast<long>(sgrid) * kThreads * V));
+    }
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
+      hipLaunchKernelGGL((bn_stats_nhwc_reg_k<T, V>), dim3(sgrid),
+                         dim3(kThreads), 0, stream, static_cast<const T*>(x),
+                         ws, C, numel);
+    } else {
 #define BF_BN_STATS(NH, VV)                                                    \
   hipLaunchKernelGGL((bn_stats_k<T, NH, VV>), dim3(sgrid), dim3(kThreads),     \
                      lds, stream, static_cast<const T*>(x), ws, C, HW, numel)
-    if (nhwc) { if (vec) BF_BN_STATS(true, V); else BF_BN_STATS(true, 1); }
-    else      { if (vec) BF_BN_STATS(false, V); else BF_BN_STATS(false, 1); }
+      if (nhwc) { if (vec) BF_BN_STATS(true, V); else BF_BN_STATS(true, 1); }
+      else      { if (vec) BF_BN_STATS(false, V); else BF_BN_STATS(false, 1); }
 #undef BF_BN_STATS
+    }
     BF_CHECK_HIP(hipGetLastError());
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_fwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
@@ -889,16 +981,27 @@ struct BnBwdLauncher {
                         hipStream_t stream) {
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(x, numel) && vec_ok<T>(dy, numel);
-    const int sgrid = bn_grid(numel / (vec ? V : 1));
+    int sgrid = bn_grid(numel / (vec ? V : 1));
+    const bool reg = nhwc && vec && (C % V == 0);
+    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C != 0)
+      sgrid *= C / static_cast<int>(std::gcd(static_cast<long>(C),
+                                             static_cast<long>(sgrid) * kThreads * V));
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
+      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_reg_k<T, V>), dim3(sgrid),
+                         dim3(kThreads), 0, stream, static_cast<const T*>(x),
+                         static_cast<const T*>(dy), gamma, beta, save_mean,
+                         save_invstd, ws, C, numel);
+    } else {
 #define BF_BN_RED(NH, VV)                                                      \
   hipLaunchKernelGGL((bn_bwd_reduce_k<T, NH, VV>), dim3(sgrid),                \
                      dim3(kThreads), lds, stream, static_cast<const T*>(x),    \
                      static_cast<const T*>(dy), gamma, beta, save_mean,        \
                      save_invstd, ws, C, HW, numel)
-    if (nhwc) { if (vec) BF_BN_RED(true, V); else BF_BN_RED(true, 1); }
-    else      { if (vec) BF_BN_RED(false, V); else BF_BN_RED(false, 1); }
+      if (nhwc) { if (vec) BF_BN_RED(true, V); else BF_BN_RED(true, 1); }
+      else      { if (vec) BF_BN_RED(false, V); else BF_BN_RED(false, 1); }
 #undef BF_BN_RED
+    }
     BF_CHECK_HIP(hipGetLastError());
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_bwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
