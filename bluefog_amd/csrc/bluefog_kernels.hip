@@ -434,11 +434,22 @@ __global__ __launch_bounds__(kThreads) void bn_stats_nhwc_reg_k(
       sq[v] += val * val;
     }
   }
+  // workgroup LDS reduction of the per-lane totals, then one global
+  // atomic per channel per workgroup (global contention ~ gridDim, not
+  // ~ elements/threads)
+  extern __shared__ float lds[];  // [2C]
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int v = 0; v < VEC; ++v) {
     const int c = (c0 + v) % C;
-    if (sum[v] != 0.f) atomicAdd(&ws[c], sum[v]);
-    if (sq[v] != 0.f) atomicAdd(&ws[C + c], sq[v]);
+    if (sum[v] != 0.f) atomicAdd(&lds[c], sum[v]);
+    if (sq[v] != 0.f) atomicAdd(&lds[C + c], sq[v]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lds[c] != 0.f) atomicAdd(&ws[c], lds[c]);
+    if (lds[C + c] != 0.f) atomicAdd(&ws[C + c], lds[C + c]);
   }
 }
 
@@ -477,11 +488,19 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_nhwc_reg_k(
       }
     }
   }
+  extern __shared__ float lds[];  // [2C]
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int v = 0; v < VEC; ++v) {
     const int c = (c0 + v) % C;
-    if (sg[v] != 0.f) atomicAdd(&ws[c], sg[v]);
-    if (sgx[v] != 0.f) atomicAdd(&ws[C + c], sgx[v]);
+    if (sg[v] != 0.f) atomicAdd(&lds[c], sg[v]);
+    if (sgx[v] != 0.f) atomicAdd(&lds[C + c], sgx[v]);
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    if (lds[c] != 0.f) atomicAdd(&ws[c], lds[c]);
+    if (lds[C + c] != 0.f) atomicAdd(&ws[C + c], lds[C + c]);
   }
 }
 
@@ -941,7 +960,7 @@ struct BnFwdLauncher {
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
     if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
       hipLaunchKernelGGL((bn_stats_nhwc_reg_k<T, V>), dim3(sgrid),
-                         dim3(kThreads), 0, stream, static_cast<const T*>(x),
+                         dim3(kThreads), lds, stream, static_cast<const T*>(x),
                          ws, C, numel);
     } else {
 #define BF_BN_STATS(NH, VV)                                                    \
@@ -989,7 +1008,7 @@ struct BnBwdLauncher {
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
     if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_reg_k<T, V>), dim3(sgrid),
-                         dim3(kThreads), 0, stream, static_cast<const T*>(x),
+                         dim3(kThreads), lds, stream, static_cast<const T*>(x),
                          static_cast<const T*>(dy), gamma, beta, save_mean,
                          save_invstd, ws, C, numel);
     } else {
