@@ -189,7 +189,8 @@ def weighted_combine_sgd(
         g = g.add(param, alpha=weight_decay)
     if momentum_buf is not None and momentum != 0:
         momentum_buf.mul_(momentum).add_(g, alpha=1.0 - dampening)
-        g = grad.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
+        # nesterov uses the weight-decay-adjusted gradient (torch semantics)
+        g = g.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
     param.add_(g, alpha=-lr)
 
 
