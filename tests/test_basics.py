@@ -185,3 +185,28 @@ def test_stall_watchdog_warns(monkeypatch, caplog):
         "neighbor_allreduce.stalled_param" in r.message and "stalled" in r.message
         for r in caplog.records
     )
+
+
+def w_allreduce_parameters():
+    import torch
+
+    import bluefog_amd as bf
+
+    bf.init()
+    rank = bf.rank()
+    params = {
+        "w": torch.ones(4, 3) * (rank + 1.0),
+        "b": torch.ones(2) * (10.0 * rank),
+    }
+    bf.allreduce_parameters(params)
+    n = bf.size()
+    expect_w = sum(r + 1.0 for r in range(n)) / n
+    expect_b = sum(10.0 * r for r in range(n)) / n
+    assert torch.allclose(params["w"], torch.full((4, 3), expect_w))
+    assert torch.allclose(params["b"], torch.full((2,), expect_b))
+
+
+def test_allreduce_parameters():
+    from tests.util import run_dist
+
+    run_dist(w_allreduce_parameters, 3)
