@@ -210,3 +210,23 @@ def test_allreduce_parameters():
     from tests.util import run_dist
 
     run_dist(w_allreduce_parameters, 3)
+
+
+def w_shutdown_clean():
+    import torch
+
+    import bluefog_amd as bf
+
+    bf.init()
+    t = torch.ones(8) * bf.rank()
+    bf.allreduce(t, name="pre_shutdown")
+    bf.win_create(t, "sw")
+    bf.barrier()
+    bf.shutdown()
+    assert not bf._ctx().is_initialized()
+
+
+def test_shutdown_clean():
+    from tests.util import run_dist
+
+    run_dist(w_shutdown_clean, 2)
