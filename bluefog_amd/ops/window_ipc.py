@@ -61,7 +61,6 @@ class IpcWindowPeers:
         need = [f"win/{name}/ipc/{d}" for d in win.out_ranks]
         need += [f"win/{name}/ipcself/{s}" for s in win.in_ranks]
         store.wait(need, timeout_s=120.0)
-        from bluefog_amd.parallel.topology import GetRecvWeights  # noqa: F401
 
         topo = c.load_topology()
         self._peer_slot: Dict[int, torch.Tensor] = {}

@@ -568,7 +568,12 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
                 elif handle is not None:
                     output = bf.synchronize(handle)
                     b["flat"].copy_(output)
-            self._reduce_delay = self._num_steps_per_communication
+            if self._handles:
+                # reset the local-step countdown only when communication
+                # actually fired this round (reference optimizers.py:437-446
+                # resets per-param only for params with handles) — so the
+                # "step every iteration, communicate every N" pattern works
+                self._reduce_delay = self._num_steps_per_communication
         self._handles.clear()
         self._synchronized = True
 

@@ -146,6 +146,33 @@ def w_awc_local_steps():
     _assert_converged(bf, model, problem, 1e-2)
 
 
+def w_awc_local_steps_stepwise():
+    """Pattern (b): step() every iteration, communication fires every
+    N-th forward (reference semantics: delay resets only on comm rounds)."""
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.set_topology(bf.ExponentialTwoGraph(bf.size()))
+    problem = _Problem(bf.rank())
+    model = _make_model()
+    opt = bf.DistributedAdaptWithCombineOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.05),
+        model=model,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+        num_steps_per_communication=2,
+    )
+    comms = 0
+    for it in range(200):
+        opt.zero_grad()
+        loss = ((model(problem.A) - problem.b) ** 2).mean()
+        loss.backward()
+        if opt._handles:
+            comms += 1
+        opt.step()
+    assert comms >= 90, f"communication fired only {comms} times in 200 iters"
+    _assert_converged(bf, model, problem, 1e-2)
+
+
 def w_atc_sgd():
     import bluefog_amd as bf
 
@@ -295,6 +322,10 @@ def test_awc_allreduce():
 
 def test_awc_local_steps():
     run_dist(w_awc_local_steps, 2, timeout=300)
+
+
+def test_awc_local_steps_stepwise():
+    run_dist(w_awc_local_steps_stepwise, 2, timeout=300)
 
 
 def test_atc_sgd():
