@@ -279,7 +279,11 @@ class _DistributedOptimizer(torch.optim.Optimizer):
     def synchronize(self):
         bf = _bf()
         with torch.no_grad():
-            if bf.size() > 1:
+            # only act when a communication round is underway (some bucket
+            # fired); with backward_passes_per_step=N the intermediate
+            # steps must neither flush nor reset the countdowns (reference
+            # optimizers.py:233-248 touches only params with handles)
+            if bf.size() > 1 and self._handles:
                 # flush buckets whose parameters never all fired (frozen /
                 # unused params): the collective must still run on every rank
                 for b in self._buckets.buckets:
@@ -287,9 +291,9 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                         self._fire_bucket(b)
                 for name, handle in self._handles.items():
                     bf.synchronize(handle)
-            for p in self._allreduce_delay:
-                self._allreduce_delay[p] = self._backward_passes_per_step
-        self._pending.clear()
+                for p in self._allreduce_delay:
+                    self._allreduce_delay[p] = self._backward_passes_per_step
+                self._pending.clear()
         self._handles.clear()
         self._synchronized = True
 
@@ -799,6 +803,7 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
                     self._error_encountered = True
             self._reduce_delay[p] -= 1
             if self._reduce_delay[p] == 0:
+                self._round_fired = True
                 with torch.no_grad():
                     if self._fused is not None and self._step_func is None:
                         b = self._bucket_of.get(p)
@@ -993,9 +998,14 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
                 elif handle is not None:
                     output = bf.synchronize(handle)
                     key.set_(output)
-            for p in self._reduce_delay:
-                self._reduce_delay[p] = self._backward_passes_per_step
-        self._pending.clear()
+            # countdowns restart only after a round in which the hooks
+            # applied updates (covers both the fused size-1 case, which has
+            # no handles, and the communicate-every-N pattern)
+            if getattr(self, "_round_fired", False):
+                for p in self._reduce_delay:
+                    self._reduce_delay[p] = self._backward_passes_per_step
+                self._pending.clear()
+            self._round_fired = False
         self._handles.clear()
         self._synchronized = True
 

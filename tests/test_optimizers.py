@@ -173,6 +173,61 @@ def w_awc_local_steps_stepwise():
     _assert_converged(bf, model, problem, 1e-2)
 
 
+def w_gradient_allreduce_accumulation():
+    """backward_passes_per_step=2 with step() every iteration: grads
+    accumulate locally, the allreduce fires every second backward."""
+    import bluefog_amd as bf
+
+    bf.init()
+    problem = _Problem(bf.rank())
+    model = _make_model()
+    opt = bf.DistributedGradientAllreduceOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.02), model=model,
+        backward_passes_per_step=2,
+    )
+    comms = 0
+    for it in range(200):
+        if it % 2 == 0:
+            opt.zero_grad()
+        loss = ((model(problem.A) - problem.b) ** 2).mean()
+        loss.backward()
+        if opt._handles:
+            comms += 1
+            opt.step()
+        else:
+            # intermediate accumulation pass: no comm, no step
+            with opt.skip_synchronize():
+                pass
+    assert 90 <= comms <= 110, comms
+    _assert_converged(bf, model, problem, 5e-2)
+
+
+def w_atc_local_steps_stepwise():
+    """ATC with backward_passes_per_step=2, step() every iteration."""
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.set_topology(bf.ExponentialTwoGraph(bf.size()))
+    problem = _Problem(bf.rank())
+    model = _make_model()
+    opt = bf.DistributedAdaptThenCombineOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.05),
+        model=model,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+        backward_passes_per_step=2,
+    )
+    fired = 0
+    for it in range(200):
+        opt.zero_grad()
+        loss = ((model(problem.A) - problem.b) ** 2).mean()
+        loss.backward()
+        if opt._handles:
+            fired += 1
+        opt.step()
+    assert fired >= 90, fired
+    _assert_converged(bf, model, problem, 1e-2)
+
+
 def w_atc_sgd():
     import bluefog_amd as bf
 
@@ -326,6 +381,14 @@ def test_awc_local_steps():
 
 def test_awc_local_steps_stepwise():
     run_dist(w_awc_local_steps_stepwise, 2, timeout=300)
+
+
+def test_gradient_allreduce_accumulation():
+    run_dist(w_gradient_allreduce_accumulation, 2, timeout=300)
+
+
+def test_atc_local_steps_stepwise():
+    run_dist(w_atc_local_steps_stepwise, 2, timeout=300)
 
 
 def test_atc_sgd():
