@@ -32,6 +32,12 @@ def _bool_env(name: str, default: bool) -> bool:
 
 
 def fusion_threshold_bytes() -> int:
+    # BLUEFOG_FUSION_THRESHOLD is the reference's knob name
+    # (operations.cc:474-487); BLUEFOG_BUCKET_BYTES is the alias used in
+    # this framework's docs — either works
+    v = os.environ.get("BLUEFOG_BUCKET_BYTES")
+    if v:
+        return int(v)
     return _int_env("BLUEFOG_FUSION_THRESHOLD", 64 * 1024 * 1024)
 
 
