@@ -10,10 +10,10 @@ Data-plane knobs:
 - ``BLUEFOG_ALLOW_TORCH_FALLBACK``: if "1", GPU post-ops may silently fall
   back to torch elementwise chains when the HIP extension is missing.
   Default off: on a ROCm box the native kernels must load or ops raise.
-- ``BLUEFOG_OPS_ON_CPU``: if "1", stage GPU tensors through CPU for the
-  two-sided ops (debug aid; reference mpi_ops.cc:48-50).
 - ``BLUEFOG_WIN_ON_GPU``: if "0", window buffers for GPU tensors are kept on
-  CPU (debug aid; reference mpi_win_ops.cc:52-54).
+  CPU (debug aid; reference mpi_win_ops.cc:52-54). The reference's
+  ``BLUEFOG_OPS_ON_CPU`` staging knob is intentionally not carried over:
+  ops run where the tensor lives (docs/env_variables.md).
 """
 
 import os
@@ -43,10 +43,6 @@ def fusion_threshold_bytes() -> int:
 
 def allow_torch_fallback() -> bool:
     return _bool_env("BLUEFOG_ALLOW_TORCH_FALLBACK", False)
-
-
-def ops_on_cpu() -> bool:
-    return _bool_env("BLUEFOG_OPS_ON_CPU", False)
 
 
 def win_on_gpu() -> bool:
