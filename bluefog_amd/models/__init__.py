@@ -11,3 +11,4 @@ from bluefog_amd.models.resnet import (  # noqa: F401
     resnet152,
 )
 from bluefog_amd.models.bert import BertConfig, BertForMaskedLM, bert_base  # noqa: F401
+from bluefog_amd.models.vgg import VGG, vgg11, vgg13, vgg16, vgg19  # noqa: F401,E402
