@@ -52,3 +52,29 @@ def test_interactive_error_propagates(cluster):
     # cluster still usable afterwards
     assert c.ping() == ["0", "1"]
     c.close()
+
+
+def test_ibfrun_cli_start_stop(tmp_path, monkeypatch):
+    """The ibfrun CLI itself: start 2 workers, connect, stop."""
+    import subprocess
+
+    env = dict(os.environ, BLUEFOG_IBFRUN_DIR=str(tmp_path))
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [os.path.join(root, "ibfrun"), "start", "-np", "2",
+         "--ipython-profile", "cli"],
+        capture_output=True, text=True, timeout=120, env=env, cwd=root,
+    )
+    assert r.returncode == 0, r.stderr
+    assert "started 2 workers" in r.stdout
+    try:
+        monkeypatch.setattr(ib, "_STATE_DIR", str(tmp_path))
+        c = ib.InteractiveClient(profile="cli")
+        assert c.ping() == ["0", "1"]
+        c.close()
+    finally:
+        r = subprocess.run(
+            [os.path.join(root, "ibfrun"), "stop", "--ipython-profile", "cli"],
+            capture_output=True, text=True, timeout=60, env=env, cwd=root,
+        )
+        assert r.returncode == 0, r.stderr
