@@ -226,6 +226,10 @@ def start_cluster(np_: int, profile: str = "bluefog", extra_env=None) -> dict:
             MASTER_PORT=str(master_port),
         )
         env.update(extra_env or {})
+        # give each worker its own log file: inheriting the launcher's
+        # stdout/stderr pipes would keep them open past `ibfrun start`,
+        # hanging any caller that waits for EOF
+        log = open(os.path.join(_STATE_DIR, f"ibfrun_{profile}_worker{rank}.log"), "ab")
         proc = subprocess.Popen(
             [
                 sys.executable,
@@ -235,7 +239,10 @@ def start_cluster(np_: int, profile: str = "bluefog", extra_env=None) -> dict:
             ],
             env=env,
             start_new_session=True,
+            stdout=log,
+            stderr=subprocess.STDOUT,
         )
+        log.close()
         pids.append(proc.pid)
     state = {
         "np": np_,
