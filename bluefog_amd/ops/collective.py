@@ -38,6 +38,7 @@ def _allreduce_impl(
     is_hierarchical_local: bool,
     name: Optional[str],
 ) -> int:
+    engine.wait_if_suspended()
     c = ctx()
     if is_hierarchical_local:
         assert c.is_homogeneous(), (

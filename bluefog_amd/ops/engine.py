@@ -36,6 +36,17 @@ def auto_name(op: str, name: Optional[str]) -> str:
     return f"{op}.noname.{c}"
 
 
+def wait_if_suspended() -> None:
+    """Honor bf.suspend()/resume(): while suspended, op submission blocks
+    (reference analog: suspending the background comm thread,
+    operations.cc:1392-1400 — ops queue until resume)."""
+    c = ctx()
+    import time as _time
+
+    while c.suspended:
+        _time.sleep(0.001)
+
+
 def submit(
     name: str,
     works: List,

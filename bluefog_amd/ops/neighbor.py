@@ -83,6 +83,7 @@ def post_neighbor_exchange_raw(
     average+step path): resolve weights, post the batched RCCL exchange and
     return (works, gathered, src_weight_list, self_weight, keep_alive)
     WITHOUT scheduling any post-processing — the caller fuses its own."""
+    engine.wait_if_suspended()
     tensor = tensor.detach()
     if not tensor.is_contiguous():
         tensor = tensor.contiguous()
@@ -112,6 +113,7 @@ def _neighbor_allreduce_nonblocking_impl(
     enable_topo_check: bool,
     name: Optional[str],
 ) -> int:
+    engine.wait_if_suspended()
     c = ctx()
     tensor = tensor.detach()
     if not tensor.is_contiguous():

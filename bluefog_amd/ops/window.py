@@ -443,6 +443,9 @@ def win_update_then_collect(name: str, require_mutex: bool = True) -> torch.Tens
 
 
 def _submit_win_op(name: str, op_label: str, job) -> int:
+    from bluefog_amd.ops import engine
+
+    engine.wait_if_suspended()
     h = handle_manager().allocate(f"{op_label}.{name}.{_op_seq()}")
     timeline().start_activity(name, op_label.upper())
     h.future = registry().executor().submit(job)

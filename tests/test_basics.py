@@ -230,3 +230,41 @@ def test_shutdown_clean():
     from tests.util import run_dist
 
     run_dist(w_shutdown_clean, 2)
+
+
+def w_suspend_blocks_ops():
+    """bf.suspend() must hold back new op submission until resume
+    (reference: suspending the background comm thread)."""
+    import threading
+    import time as _time
+
+    import torch
+
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.suspend()
+    state = {"submitted": False}
+
+    def submitter():
+        t = torch.ones(4) * bf.rank()
+        h = bf.allreduce_nonblocking(t, name="suspended_op")
+        state["submitted"] = True
+        state["out"] = bf.synchronize(h)
+
+    th = threading.Thread(target=submitter)
+    th.start()
+    _time.sleep(0.3)
+    assert not state["submitted"], "op went through while suspended"
+    bf.resume()
+    th.join(timeout=30)
+    assert state["submitted"]
+    n = bf.size()
+    expected = sum(range(n)) / n
+    assert torch.allclose(state["out"], torch.full((4,), expected))
+
+
+def test_suspend_blocks_ops():
+    from tests.util import run_dist
+
+    run_dist(w_suspend_blocks_ops, 2)
