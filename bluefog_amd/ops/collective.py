@@ -118,6 +118,7 @@ def broadcast(
 def broadcast_nonblocking(
     tensor: torch.Tensor, root_rank: int, name: Optional[str] = None
 ) -> int:
+    engine.wait_if_suspended()
     output = tensor.detach().clone()
     work = dist.broadcast(output, src=root_rank, async_op=True)
     return engine.submit(
@@ -137,6 +138,7 @@ def broadcast_(tensor: torch.Tensor, root_rank: int, name: Optional[str] = None)
 def broadcast_nonblocking_(
     tensor: torch.Tensor, root_rank: int, name: Optional[str] = None
 ) -> int:
+    engine.wait_if_suspended()
     work = dist.broadcast(tensor, src=root_rank, async_op=True)
     return engine.submit(
         engine.auto_name("broadcast", name),
@@ -155,6 +157,7 @@ def allgather(tensor: torch.Tensor, name: Optional[str] = None) -> torch.Tensor:
 def allgather_nonblocking(tensor: torch.Tensor, name: Optional[str] = None) -> int:
     """Concatenate the tensor from every rank along dim 0; first dims may
     differ (reference allows ragged first dims, MPI_Allgatherv)."""
+    engine.wait_if_suspended()
     c = ctx()
     size = c.size()
     t = tensor.detach()

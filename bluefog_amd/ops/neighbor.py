@@ -326,6 +326,7 @@ def hierarchical_neighbor_allreduce_nonblocking(
 ) -> int:
     import numpy as np
 
+    engine.wait_if_suspended()
     c = ctx()
     assert c.is_homogeneous(), (
         "hierarchical_neighbor_allreduce should be used under homogeneous "
