@@ -234,6 +234,7 @@ def neighbor_allgather_nonblocking(
     enable_topo_check: bool = True,
     name: Optional[str] = None,
 ) -> int:
+    engine.wait_if_suspended()
     c = ctx()
     if (src_ranks is None) != (dst_ranks is None):
         raise ValueError(
@@ -479,6 +480,7 @@ def pair_gossip_nonblocking(
     pair_weight: Optional[float] = None,
     name: Optional[str] = None,
 ) -> int:
+    engine.wait_if_suspended()
     if (self_weight is None) != (pair_weight is None):
         raise ValueError(
             "Arguments self_weight and pair_weight have to be presented at the same time"
