@@ -55,3 +55,28 @@ def test_bench_default_metric_is_baseline_config():
     assert args.batch_size == 64
     assert args.dist_optimizer == "neighbor_allreduce"
     assert args.dtype == "fp32"
+
+
+def test_bench_hierarchical_faked_machines():
+    """BASELINE config 5's code path end to end: 4 ranks, 2 faked machines
+    (BLUEFOG_NODES_PER_MACHINE), hierarchical_neighbor_allreduce."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = dict(os.environ, BLUEFOG_NODES_PER_MACHINE="2")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(_ROOT, "bench.py"), "--gpus", "4", "--steps", "2",
+         "--warmup", "1", "--model", "resnet18", "--batch-size", "2",
+         "--device", "cpu", "--dist-optimizer", "hierarchical_neighbor_allreduce"],
+        cwd=_ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:] + out.stdout[-500:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert "hierarchical" in rec["config"]["parallelism"], rec["config"]
