@@ -31,10 +31,21 @@ def init(topology_fn=None, is_weighted: bool = False):
     from bluefog_amd.utils.timeline import maybe_init_from_env
 
     maybe_init_from_env(_ctx().rank())
+    from bluefog_amd.utils import metrics as _metrics
+
+    _metrics.maybe_start_from_env(_ctx().rank())
     # make the window registry known to the context for topology guards
     from bluefog_amd.ops import window as _window
 
     _window.registry()
+
+
+def start_metrics_server(port: int):
+    """Serve Prometheus metrics for this rank on ``port`` (see
+    bluefog_amd/utils/metrics.py)."""
+    from bluefog_amd.utils import metrics as _metrics
+
+    _metrics.start_server(port)
 
 
 def shutdown():

@@ -19,6 +19,7 @@ from typing import Callable, List, Optional
 
 import torch
 
+from bluefog_amd.utils import metrics
 from bluefog_amd.utils.logging import get_logger
 from bluefog_amd.utils.timeline import timeline
 
@@ -127,6 +128,7 @@ class HandleManager:
                     self._stall_reported.add(hid)
             for hid, name, age in stalled:
                 if not self._handles.get(hid, OpHandle(-1, "")).poll():
+                    metrics.record_stall(name)
                     get_logger().warning(
                         "op %r (handle %d) has not completed for %.0f s — "
                         "one or more peer ranks likely never submitted the "
@@ -151,7 +153,8 @@ class HandleManager:
             self._outstanding_names.add(name)
             self._birth[hid] = time.monotonic()
             self._ensure_watchdog()
-            return h
+        metrics.record_submit(name)
+        return h
 
     def get(self, hid: int) -> OpHandle:
         with self._lock:
@@ -174,6 +177,11 @@ class HandleManager:
     def synchronize(self, hid: int) -> torch.Tensor:
         h = self.get(hid)
         out = h.synchronize()
+        if metrics.enabled():
+            with self._lock:
+                birth = self._birth.get(hid)
+            if birth is not None:
+                metrics.record_latency(h.name, time.monotonic() - birth)
         self.release(hid)
         return out
 
