@@ -57,3 +57,9 @@ def test_example_resnet_tiny():
         "--val-samples", "16", "--batch-size", "8", "--no-checkpoint",
     )
     assert "epoch 0" in out
+
+
+def test_example_average_consensus_async():
+    out = _run_example(2, "pytorch_average_consensus.py", "--asynchronous-mode",
+                       "--data-size", "1000")
+    assert "consensus reached" in out
