@@ -1,7 +1,7 @@
 # Copyright 2026. Licensed under the Apache License, Version 2.0.
 """Dispatch layer over the native HIP extension ``bluefog_amd._C``.
 
-The extension carries the hand-written CDNA4 kernels (csrc/kernels.hip) that
+The extension carries the hand-written CDNA4 kernels (csrc/bluefog_kernels.hip) that
 replace the reference's chain of torch slice arithmetic in the
 post-communication callbacks (reference: bluefog/torch/mpi_ops.cc:99-164,
 mpi_win_ops.cc:185-279) and the reference's lone CUDA kernel family

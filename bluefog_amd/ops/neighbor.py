@@ -10,7 +10,7 @@ one-peer Exponential-2 schedule each iteration's single send+recv pair
 lands on a distinct xGMI link (7 per GPU, ~153 GB/s each) with zero
 contention — the reason this beats ring allreduce on this fabric. The
 post-communication weighted average is one hand-written CDNA4 kernel
-(csrc/kernels.hip) on a side stream instead of the reference's chain of
+(csrc/bluefog_kernels.hip) on a side stream instead of the reference's chain of
 torch slice ops (mpi_ops.cc:99-164).
 """
 
