@@ -411,3 +411,67 @@ def test_neighbor_allreduce_dims():
 
 def test_inner_outer_dynamic():
     run_dist(w_inner_outer_dynamic, 8)
+
+
+def w_weighted_topology_default_resolution():
+    """set_topology(is_weighted=True): neighbor_allreduce with no explicit
+    weights must use the graph's Metropolis-Hastings weights."""
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    topo = tu.MeshGrid2DGraph(bf.size())
+    bf.set_topology(topo, is_weighted=True)
+    rank = bf.rank()
+    t = torch.ones(6, dtype=torch.float64) * (rank + 1)
+    out = bf.neighbor_allreduce(t)
+    self_w, nbr_w = tu.GetRecvWeights(topo, rank)
+    expected = self_w * (rank + 1) + sum(w * (r + 1) for r, w in nbr_w.items())
+    assert torch.allclose(out, torch.full_like(t, expected)), (
+        rank, float(out[0]), expected)
+
+
+def w_error_paths():
+    import bluefog_amd as bf
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    t = torch.ones(3)
+    # dst_weights containing self is rejected
+    try:
+        bf.neighbor_allreduce(
+            t, self_weight=0.5, src_weights={(rank + 1) % size: 0.5},
+            dst_weights=[rank],
+        )
+        raise AssertionError("self in dst_weights must be rejected")
+    except ValueError:
+        pass
+    # src_weights must be a dict
+    try:
+        bf.neighbor_allreduce(t, self_weight=0.5, src_weights=[0], dst_weights=[0])
+        raise AssertionError("non-dict src_weights must be rejected")
+    except ValueError:
+        pass
+    # unknown window name
+    try:
+        bf.win_update("never_created")
+        raise AssertionError("unknown window must raise")
+    except ValueError:
+        pass
+    # duplicated window name
+    assert bf.win_create(t, "dupwin")
+    try:
+        bf.win_create(t, "dupwin")
+        raise AssertionError("duplicate window name must raise")
+    except ValueError:
+        pass
+    bf.win_free("dupwin")
+    bf.barrier()
+
+
+def test_weighted_topology_default_resolution():
+    run_dist(w_weighted_topology_default_resolution, 4)
+
+
+def test_error_paths():
+    run_dist(w_error_paths, 2)
