@@ -176,6 +176,11 @@ def resolve_recv_weights(
             dst_weights_d = {int(d): 1.0 for d in dst_weights}
         else:
             dst_weights_d = {int(d): float(w) for d, w in dst_weights.items()}
+        if c.rank() in dst_weights_d:
+            raise ValueError(
+                "The key of dst_weights should only contain other ranks "
+                "(self-rank is not allowed; use self_weight)."
+            )
         dst_weighting_enabled = not np.allclose(list(dst_weights_d.values()), 1.0)
 
     if self_weight is None and src_weights is None:
@@ -197,6 +202,11 @@ def resolve_recv_weights(
             )
         if not isinstance(self_weight, float):
             raise ValueError("Argument self_weight has to be a float for self rank.")
+        if c.rank() in src_weights:
+            raise ValueError(
+                "The key of src_weights should only contain other ranks "
+                "(self-rank is not allowed; use self_weight)."
+            )
         if not dynamic_enabled and not set(src_weights.keys()).issubset(
             set(c.in_neighbor_ranks())
         ):
