@@ -373,6 +373,11 @@ def hierarchical_neighbor_allreduce_nonblocking(
             )
         if isinstance(dst_machine_weights, (list, tuple)):
             dst_machine_weights = {int(d): 1.0 for d in dst_machine_weights}
+        if c.machine_rank() in dst_machine_weights or c.machine_rank() in src_machine_weights:
+            raise ValueError(
+                "src/dst_machine_weights should only contain other machines "
+                "(self machine is not allowed; use self_weight)."
+            )
         dst_weighting_enabled = not np.allclose(list(dst_machine_weights.values()), 1.0)
     else:
         raise ValueError(
