@@ -241,6 +241,11 @@ def neighbor_allgather_nonblocking(
             "Argument dst_ranks and src_ranks have to be presented at the same time"
         )
     dynamic = src_ranks is not None
+    if dynamic and (c.rank() in src_ranks or c.rank() in dst_ranks):
+        raise ValueError(
+            "src_ranks/dst_ranks should only contain other ranks "
+            "(self-rank is not allowed)."
+        )
     if not dynamic:
         src_ranks = c.in_neighbor_ranks()
         dst_ranks = c.out_neighbor_ranks()
@@ -486,6 +491,8 @@ def pair_gossip_nonblocking(
     name: Optional[str] = None,
 ) -> int:
     engine.wait_if_suspended()
+    if target_rank == ctx().rank():
+        raise ValueError("pair_gossip target_rank must be another rank.")
     if (self_weight is None) != (pair_weight is None):
         raise ValueError(
             "Arguments self_weight and pair_weight have to be presented at the same time"
