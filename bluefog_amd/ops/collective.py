@@ -64,6 +64,7 @@ def _allreduce_impl(
         finalize,
         output.device,
         keep_alive=(output,),
+        nbytes=output.numel() * output.element_size(),
     )
 
 

@@ -53,10 +53,14 @@ def submit(
     finalize: Callable[[], torch.Tensor],
     device: torch.device,
     keep_alive: Sequence[torch.Tensor] = (),
+    nbytes: Optional[int] = None,
 ) -> int:
     """Create a handle for posted works + a post-op callback and, on GPU,
     run the callback now on the side stream (stream-ordered)."""
     h = handle_manager().allocate(name)
+    from bluefog_amd.utils import metrics
+
+    metrics.record_submit(name, nbytes)
     timeline().start_activity(name, "COMMUNICATE")
     if device.type == "cuda":
         side = ctx().side_stream()

@@ -153,7 +153,6 @@ class HandleManager:
             self._outstanding_names.add(name)
             self._birth[hid] = time.monotonic()
             self._ensure_watchdog()
-        metrics.record_submit(name)
         return h
 
     def get(self, hid: int) -> OpHandle:

@@ -146,6 +146,7 @@ def _neighbor_allreduce_nonblocking_impl(
         finalize,
         tensor.device,
         keep_alive=keep_alive + [output],
+        nbytes=tensor.numel() * tensor.element_size() * (1 + len(src_ranks)),
     )
 
 

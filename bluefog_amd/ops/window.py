@@ -442,11 +442,13 @@ def win_update_then_collect(name: str, require_mutex: bool = True) -> torch.Tens
 # ---------------------------------------------------------------------------
 
 
-def _submit_win_op(name: str, op_label: str, job) -> int:
+def _submit_win_op(name: str, op_label: str, job, nbytes=None) -> int:
     from bluefog_amd.ops import engine
+    from bluefog_amd.utils import metrics
 
     engine.wait_if_suspended()
     h = handle_manager().allocate(f"{op_label}.{name}.{_op_seq()}")
+    metrics.record_submit(f"{op_label}.{name}", nbytes)
     timeline().start_activity(name, op_label.upper())
     h.future = registry().executor().submit(job)
     return h.id
