@@ -944,6 +944,7 @@ struct BnFwdLauncher {
                         float* run_var, float* save_mean, float* save_invstd,
                         float* ws, float momentum, float eps, int C, long HW,
                         long numel, bool nhwc, hipStream_t stream) {
+    if (C > kMaxBnChannels) return hipErrorInvalidValue;  // LDS budget
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(x, numel);
     int sgrid = bn_grid(numel / (vec ? V : 1));
@@ -998,6 +999,7 @@ struct BnBwdLauncher {
                         float* ws, float* dgamma, float* dbeta, float* coef,
                         int C, long HW, long numel, bool nhwc,
                         hipStream_t stream) {
+    if (C > kMaxBnChannels) return hipErrorInvalidValue;  // LDS budget
     constexpr int V = vec_width<T>();
     const bool vec = vec_ok<T>(x, numel) && vec_ok<T>(dy, numel);
     int sgrid = bn_grid(numel / (vec ? V : 1));
