@@ -103,6 +103,19 @@ def run(numel, iters, dtype, dev):
         )
         add("combine_adam[1nbr]", ms, 8 * numel * esz)
 
+    if dtype in (torch.float32, torch.bfloat16):
+        # fused residual add+ReLU (ResNet block join shape, channels_last)
+        a4 = torch.randn(64, 256, 56, 56, device=dev).to(dtype).to(
+            memory_format=torch.channels_last)
+        b4 = torch.randn_like(a4)
+        o4 = torch.empty_like(a4)
+        g4 = torch.randn_like(a4)
+        n4 = a4.numel()
+        ms = time_kernel(lambda: _C.add_relu_fwd(o4, a4, b4), iters)
+        add("add_relu_fwd[64x256x56x56]", ms, 3 * n4 * esz)
+        ms = time_kernel(lambda: _C.relu_bwd_mask(o4, g4, a4), iters)
+        add("relu_bwd_mask[64x256x56x56]", ms, 3 * n4 * esz)
+
     if dtype == torch.float32:
         # fused BN+ReLU pipeline (opt-in path) on a ResNet50-stem-shaped
         # activation: measures how far the reductions are from peak
