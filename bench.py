@@ -164,6 +164,11 @@ def main():
         )
     elif opt_name == "win_put":
         optimizer = bf.DistributedWinPutOptimizer(base_opt, model=model)
+        if n > 1:
+            dyn_gen = iter(
+                {bf.out_neighbor_ranks()[i % len(bf.out_neighbor_ranks())]: 1.0}
+                for i in __import__("itertools").count()
+            )
     elif opt_name == "pushsum":
         optimizer = bf.DistributedPushSumOptimizer(base_opt, model=model)
     elif opt_name == "atc":
@@ -185,7 +190,13 @@ def main():
     def set_dynamic():
         if dyn_gen is None:
             return
-        send, recv = next(dyn_gen)
+        sched = next(dyn_gen)
+        if opt_name == "win_put":
+            # rotate the gossip destination, one xGMI link per iteration
+            # (reference pytorch_benchmark.py:183-186)
+            optimizer.dst_weights = sched
+            return
+        send, recv = sched
         w = 1.0 / (len(recv) + 1)
         optimizer.self_weight = w
         optimizer.src_weights = {r: w for r in recv}

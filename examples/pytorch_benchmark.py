@@ -99,13 +99,22 @@ def main():
         )
         optimizer = awc_or_atc(base_opt, model=model, communication_type=ct)
 
+    # per-iteration dynamic schedules (reference pytorch_benchmark.py:160-200)
     dyn_gen = None
-    if (
-        not args.disable_dynamic_topology
-        and bf.size() > 1
-        and name == "neighbor_allreduce"
-    ):
-        dyn_gen = tu.GetDynamicOnePeerSendRecvRanks(topo, bf.rank())
+    dyn_machine_gen = None
+    if not args.disable_dynamic_topology and bf.size() > 1:
+        if name == "neighbor_allreduce":
+            if bf.is_homogeneous() and bf.size() > bf.local_size():
+                dyn_gen = tu.GetInnerOuterExpo2DynamicSendRecvRanks(
+                    bf.size(), local_size=bf.local_size(), self_rank=bf.rank()
+                )
+            else:
+                dyn_gen = tu.GetDynamicOnePeerSendRecvRanks(topo, bf.rank())
+        elif name == "hierarchical_neighbor_allreduce" and bf.machine_size() > 1:
+            dyn_machine_gen = tu.GetExp2DynamicSendRecvMachineRanks(
+                world_size=bf.size(), local_size=bf.local_size(),
+                self_rank=bf.rank(), local_rank=bf.local_rank(),
+            )
 
     bf.broadcast_parameters(model.state_dict(), root_rank=0)
 
@@ -115,13 +124,33 @@ def main():
         data = data.to(memory_format=torch.channels_last)
     loss_fn = torch.nn.CrossEntropyLoss()
 
+    step_counter = [0]
+
     def benchmark_step():
+        it = step_counter[0]
+        step_counter[0] += 1
         if dyn_gen is not None:
             send, recv = next(dyn_gen)
             w = 1.0 / (len(recv) + 1)
             optimizer.self_weight = w
             optimizer.src_weights = {r: w for r in recv}
             optimizer.dst_weights = send
+            optimizer.enable_topo_check = False
+        elif dyn_machine_gen is not None:
+            send_m, recv_m = next(dyn_machine_gen)
+            w = 1.0 / (len(recv_m) + 1)
+            optimizer.self_weight = w
+            optimizer.src_machine_weights = {r: w for r in recv_m}
+            optimizer.dst_machine_weights = send_m
+            optimizer.enable_topo_check = False
+        elif (
+            name == "win_put"
+            and not args.disable_dynamic_topology
+            and bf.out_neighbor_ranks()
+        ):
+            # rotate the gossip destination, one out-neighbor per iteration
+            nbrs = bf.out_neighbor_ranks()
+            optimizer.dst_weights = {nbrs[it % len(nbrs)]: 1.0}
         optimizer.zero_grad()
         loss = loss_fn(model(data), target)
         loss.backward()
