@@ -132,3 +132,33 @@ def test_combine_sgd_fallback_matches_torch_sgd(numel, momentum, nesterov, seed)
     # torch SGD's first momentum step uses buf=grad (no (1-dampening) scale);
     # our kernel matches that convention
     np.testing.assert_allclose(p1.detach().numpy(), p2.detach().numpy(), atol=1e-10)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    machines=st.sampled_from([2, 4, 8]),
+    local=st.sampled_from([2, 4]),
+    rounds=st.integers(1, 4),
+)
+def test_machine_generator_mutually_consistent(machines, local, rounds):
+    """GetExp2DynamicSendRecvMachineRanks: at every iteration the machine-
+    level send/recv sets must mirror across the leader ranks."""
+    world = machines * local
+    gens = {
+        r: tu.GetExp2DynamicSendRecvMachineRanks(
+            world_size=world, local_size=local, self_rank=r, local_rank=r % local
+        )
+        for r in range(world)
+    }
+    for _ in range(rounds):
+        sends, recvs = {}, {}
+        for r, g in gens.items():
+            s_m, r_m = next(g)
+            m = r // local
+            sends[m] = set(s_m)
+            recvs[m] = set(r_m)
+        for m, dsts in sends.items():
+            for d in dsts:
+                assert m in recvs[d], (machines, local, m, d, sends, recvs)
+            for srcm in recvs[m]:
+                assert m in sends[srcm]
