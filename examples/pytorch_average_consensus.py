@@ -58,8 +58,9 @@ def main():
                 break
         bf.win_free("consensus_win")
 
+    status = "consensus reached" if err < args.atol else "max iterations hit"
     print(
-        f"[rank {bf.rank()}] consensus reached in {it + 1} iterations, "
+        f"[rank {bf.rank()}] {status} after {it + 1} iterations, "
         f"relative error {err.item():.3e}"
     )
     assert err < 1e-3, "consensus failed to converge"
