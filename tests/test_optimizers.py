@@ -291,6 +291,9 @@ def w_pull_get_optimizer():
 
 
 def w_push_sum_optimizer():
+    # NOTE: genuinely asynchronous gossip — on a heavily oversubscribed host
+    # (e.g. pytest-xdist -n 4 on a small box) the worker threads starve and
+    # convergence slows arbitrarily; run serially for a meaningful signal.
     import bluefog_amd as bf
 
     bf.init()
