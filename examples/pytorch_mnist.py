@@ -41,10 +41,13 @@ class Net(nn.Module):
 
 
 def synthetic_mnist(n, seed):
+    # class-conditional Gaussian blobs so the problem is actually learnable;
+    # the class CENTERS are shared across ranks (fixed seed) so every
+    # rank's labels mean the same thing — only the samples are per-rank
+    gc = torch.Generator().manual_seed(4242)
+    centers = torch.randn(10, 1, 28, 28, generator=gc)
     g = torch.Generator().manual_seed(seed)
-    # class-conditional Gaussian blobs so the problem is actually learnable
     y = torch.randint(0, 10, (n,), generator=g)
-    centers = torch.randn(10, 1, 28, 28, generator=g)
     x = centers[y] + 0.5 * torch.randn(n, 1, 28, 28, generator=g)
     return torch.utils.data.TensorDataset(x, y)
 
