@@ -175,13 +175,18 @@ class HandleManager:
 
     def synchronize(self, hid: int) -> torch.Tensor:
         h = self.get(hid)
-        out = h.synchronize()
-        if metrics.enabled():
-            with self._lock:
-                birth = self._birth.get(hid)
-            if birth is not None:
-                metrics.record_latency(h.name, time.monotonic() - birth)
-        self.release(hid)
+        try:
+            out = h.synchronize()
+        finally:
+            # release even when the op failed, so the name can be reused
+            # (a raised window/collective op would otherwise poison retries
+            # with DUPLICATE_NAME_ERROR)
+            if metrics.enabled():
+                with self._lock:
+                    birth = self._birth.get(hid)
+                if birth is not None:
+                    metrics.record_latency(h.name, time.monotonic() - birth)
+            self.release(hid)
         return out
 
 

@@ -268,3 +268,40 @@ def test_suspend_blocks_ops():
     from tests.util import run_dist
 
     run_dist(w_suspend_blocks_ops, 2)
+
+
+def w_failed_op_releases_name():
+    """A failing op must release its handle/name so a retry can reuse it."""
+    import torch
+
+    import bluefog_amd as bf
+    from bluefog_amd.ops import engine
+    from bluefog_amd.ops.handles import handle_manager
+
+    bf.init()
+
+    class _Boom:
+        def is_completed(self):
+            return True
+
+        def wait(self):
+            raise RuntimeError("transport exploded")
+
+    h = handle_manager().allocate("retry_me")
+    h.works = [_Boom()]
+    try:
+        engine.synchronize(h.id)
+        raise AssertionError("expected the op failure to propagate")
+    except RuntimeError:
+        pass
+    # name free again: a real op under the same name succeeds
+    t = torch.ones(3) * bf.rank()
+    out = bf.allreduce(t, name="retry_me")
+    n = bf.size()
+    assert torch.allclose(out, torch.full((3,), sum(range(n)) / n))
+
+
+def test_failed_op_releases_name():
+    from tests.util import run_dist
+
+    run_dist(w_failed_op_releases_name, 2)
