@@ -100,8 +100,10 @@ def synchronize(handle: int) -> torch.Tensor:
 def wait(handle: int) -> torch.Tensor:
     """Like synchronize, but fully blocks the host until retirement."""
     h = handle_manager().get(handle)
-    out = h.wait_host()
-    handle_manager().release(handle)
+    try:
+        out = h.wait_host()
+    finally:
+        handle_manager().release(handle)
     return out
 
 

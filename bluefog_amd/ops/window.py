@@ -693,6 +693,8 @@ def win_poll(handle: int) -> bool:
 
 def win_wait(handle: int) -> bool:
     h = handle_manager().get(handle)
-    h.synchronize()
-    handle_manager().release(handle)
+    try:
+        h.synchronize()
+    finally:
+        handle_manager().release(handle)
     return True
