@@ -109,9 +109,6 @@ class Window:
         i = self.in_ranks.index(src)
         return self.block[i].reshape(self.self_tensor.shape)
 
-    def slot_index_for(self, in_ranks: List[int], origin: int) -> int:
-        return in_ranks.index(origin)
-
 
 class WindowRegistry:
     def __init__(self):
