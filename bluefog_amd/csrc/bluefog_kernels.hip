@@ -871,6 +871,11 @@ struct CombineAdamLauncher {
     const bool vec = vec_ok<T>(p, numel) && vec_ok<T>(grad, numel) &&
                      vec_ok<float>(exp_avg, numel) &&
                      vec_ok<float>(exp_avg_sq, numel) &&
+                     // Pack<float,V> is alignas(sizeof(float)*V) — 32B when
+                     // T is 2-byte (V=8); 16B-aligned-but-not-32B state
+                     // views must take the scalar path
+                     (reinterpret_cast<uintptr_t>(exp_avg) % (sizeof(float) * V) == 0) &&
+                     (reinterpret_cast<uintptr_t>(exp_avg_sq) % (sizeof(float) * V) == 0) &&
                      (n_nbr == 0 || vec_ok<T>(gathered, numel));
     const int grid = grid_for((numel + (vec ? V : 1) - 1) / (vec ? V : 1));
 #define BF_LAUNCH_ADAM(VV)                                                     \

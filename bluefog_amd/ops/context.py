@@ -18,6 +18,7 @@ import atexit
 import datetime
 import os
 import socket
+import threading
 from typing import Callable, Dict, List, Optional
 
 import torch
@@ -56,7 +57,10 @@ class BlueFogContext:
         self._cross_group = None
         self._local_groups_built = False
         self._machine_rank_lists: List[List[int]] = []
-        self._suspended = False
+        # event semantics: set = running, cleared = suspended (submitters
+        # block in Event.wait instead of spin-polling)
+        self._running = threading.Event()
+        self._running.set()
         self._skip_negotiate = True  # program-order contract is the default
         self._owns_process_group = False
         # window registry lives in ops.window; it registers itself here so
@@ -254,14 +258,17 @@ class BlueFogContext:
         return self._device
 
     def suspend(self) -> None:
-        self._suspended = True
+        self._running.clear()
 
     def resume(self) -> None:
-        self._suspended = False
+        self._running.set()
 
     @property
     def suspended(self) -> bool:
-        return self._suspended
+        return not self._running.is_set()
+
+    def wait_until_running(self) -> None:
+        self._running.wait()
 
     def set_skip_negotiate_stage(self, value: bool) -> None:
         self._skip_negotiate = bool(value)

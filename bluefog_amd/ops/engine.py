@@ -39,12 +39,9 @@ def auto_name(op: str, name: Optional[str]) -> str:
 def wait_if_suspended() -> None:
     """Honor bf.suspend()/resume(): while suspended, op submission blocks
     (reference analog: suspending the background comm thread,
-    operations.cc:1392-1400 — ops queue until resume)."""
-    c = ctx()
-    import time as _time
-
-    while c.suspended:
-        _time.sleep(0.001)
+    operations.cc:1392-1400 — ops queue until resume). Event-backed: no
+    polling while suspended, resume wakes submitters immediately."""
+    ctx().wait_until_running()
 
 
 def submit(

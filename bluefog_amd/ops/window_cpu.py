@@ -74,11 +74,13 @@ class WindowServer:
             daemon_threads = True
             allow_reuse_address = True
 
-        self._server = Server(("0.0.0.0", 0), Handler)
-        self.port = self._server.server_address[1]
         # single-node by design; container hostnames often do not resolve
-        # (use the loopback, matching the rendezvous contract)
+        # (use the loopback, matching the rendezvous contract). Bind to the
+        # advertised host — the handler unpickles payloads, so listening on
+        # all interfaces would expose a deserialization surface for nothing.
         self.host = os.environ.get("BLUEFOG_WIN_SERVER_HOST", "127.0.0.1")
+        self._server = Server((self.host, 0), Handler)
+        self.port = self._server.server_address[1]
         self._thread = threading.Thread(target=self._server.serve_forever, daemon=True)
         self._thread.start()
 

@@ -283,7 +283,10 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             # fired); with backward_passes_per_step=N the intermediate
             # steps must neither flush nor reset the countdowns (reference
             # optimizers.py:233-248 touches only params with handles)
-            if bf.size() > 1 and self._handles:
+            # gate on _pending too: a bucket with one never-firing param
+            # (unused head) leaves _handles empty even though other params
+            # fired — their gradients must still be averaged
+            if bf.size() > 1 and (self._handles or self._pending):
                 # flush buckets whose parameters never all fired (frozen /
                 # unused params): the collective must still run on every rank
                 for b in self._buckets.buckets:
@@ -985,6 +988,14 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
         from bluefog_amd.ops import hip_ext
 
         with torch.no_grad():
+            if self._fused is not None:
+                # flush partially-fired buckets (a frozen/unused param keeps
+                # the pending set non-empty so _fused_bucket_step never ran):
+                # the fired params must still be stepped and exchanged, and
+                # the collective must run on every rank
+                for b in self._buckets.buckets:
+                    if b["name"] in self._pending and b["name"] not in self._handles:
+                        self._fused_bucket_step(b)
             for key, handle in self._handles.items():
                 if isinstance(handle, tuple) and handle and handle[0] == "fused":
                     _, works, gathered, weights, self_w, keep = handle
