@@ -57,6 +57,19 @@ class ControlStore:
     def add(self, key: str, amount: int) -> int:
         return self._store.add(self._key(key), amount)
 
+    def multi_set(self, pairs: dict) -> None:
+        """Write many keys in ONE store round-trip (values: bytes)."""
+        if not pairs:
+            return
+        self._store.multi_set([self._key(k) for k in pairs], list(pairs.values()))
+
+    def multi_get(self, keys: List[str]) -> List[bytes]:
+        """Read many keys in ONE store round-trip. Every key must already
+        exist (TCPStore blocks on missing keys)."""
+        if not keys:
+            return []
+        return self._store.multi_get([self._key(k) for k in keys])
+
     def check(self, keys: List[str]) -> bool:
         return self._store.check([self._key(k) for k in keys])
 

@@ -101,6 +101,11 @@ class Window:
             for i in range(len(in_ranks)):
                 self.block[i].copy_(tensor.reshape(shape))
         self.associated_p = 1.0
+        # local sequence numbers behind the single-writer version counters
+        # (seeded from the store at win_create; mutated only on the window
+        # worker thread)
+        self.put_seq: Dict[int, int] = {}  # dst -> #puts/accums sent there
+        self.get_seq: Dict[int, int] = {}  # src -> #gets pulled from there
         # transport attachments (set by the registry)
         self.ipc = None  # window_ipc.IpcWindowPeers
         self.cpu_addrs = None  # {rank: (host, port)} for the TCP fallback
@@ -199,8 +204,21 @@ def registry() -> WindowRegistry:
 # ---------------------------------------------------------------------------
 
 
-def _ver_key(name: str, owner: int, nbr: int) -> str:
-    return f"win/{name}/ver/{owner}/{nbr}"
+# Version bookkeeping: the update count of owner's buffer for nbr is split
+# into two single-writer counters — puts/accumulates bump the "put" key
+# (writer: the origin rank), gets bump the "get" key (writer: the owner) —
+# so each writer keeps a local sequence number and publishes ALL its
+# destinations' bumps in ONE batched multi_set round-trip instead of one
+# atomic add per destination (reference analog: the chunked-put
+# amortization of mpi_controller.cc:952-1033).
+
+
+def _ver_put_key(name: str, owner: int, nbr: int) -> str:
+    return f"win/{name}/verp/{owner}/{nbr}"
+
+
+def _ver_get_key(name: str, owner: int, nbr: int) -> str:
+    return f"win/{name}/verg/{owner}/{nbr}"
 
 
 def _ack_key(name: str, owner: int, nbr: int) -> str:
@@ -269,6 +287,17 @@ def win_create(tensor: torch.Tensor, name: str, zero_init: bool = False) -> bool
     reg.register(win)
     _p_self_set(c.store, name, c.rank(), 1.0)
     win.associated_p = 1.0
+    me = c.rank()
+    # create-or-read every version/ack counter this rank will touch so the
+    # batched multi_get/multi_set paths never block on a missing key, and
+    # seed the local sequence numbers from whatever a previous same-named
+    # window left behind (counter() is add(0): creates "0" if absent)
+    for r in win.in_ranks:
+        win.get_seq[r] = c.store.counter(_ver_get_key(name, me, r))
+        c.store.counter(_ver_put_key(name, me, r))
+        c.store.counter(_ack_key(name, me, r))
+    for dst in win.out_ranks:
+        win.put_seq[dst] = c.store.counter(_ver_put_key(name, dst, me))
 
     if use_gpu_path:
         from bluefog_amd.ops import window_ipc
@@ -313,11 +342,15 @@ def get_win_version(name: str) -> Dict[int, int]:
     c = ctx()
     win = registry().get(name)
     me = c.rank()
-    out = {}
+    keys = []
     for nbr in win.in_ranks:
-        puts = c.store.counter(_ver_key(name, me, nbr))
-        acked = c.store.counter(_ack_key(name, me, nbr))
-        out[nbr] = puts - acked
+        keys += [_ver_put_key(name, me, nbr), _ver_get_key(name, me, nbr),
+                 _ack_key(name, me, nbr)]
+    vals = c.store.multi_get(keys)  # one round-trip for all neighbors
+    out = {}
+    for i, nbr in enumerate(win.in_ranks):
+        puts, gets, acked = (int(vals[3 * i + j]) for j in range(3))
+        out[nbr] = puts + gets - acked
     return out
 
 
@@ -413,10 +446,18 @@ def win_update(
                     win.neighbor_buffer(r).zero_()
                     if _ops_with_associated_p:
                         _p_slot_set(c.store, name, me, r, 0.0)
-            # mark buffers as seen
-            for r in win.in_ranks:
-                puts = c.store.counter(_ver_key(name, me, r))
-                c.store.set(_ack_key(name, me, r), str(puts).encode())
+            # mark buffers as seen: one read + one write round-trip for
+            # ALL in-neighbors (ack = put_cnt + get_cnt at this moment)
+            if win.in_ranks:
+                keys = []
+                for r in win.in_ranks:
+                    keys += [_ver_put_key(name, me, r), _ver_get_key(name, me, r)]
+                vals = c.store.multi_get(keys)
+                acks = {}
+                for i, r in enumerate(win.in_ranks):
+                    total = int(vals[2 * i]) + int(vals[2 * i + 1])
+                    acks[_ack_key(name, me, r)] = str(total).encode()
+                c.store.multi_set(acks)
     finally:
         if require_mutex:
             c.store.mutex_release(_mutex_name(name, me))
@@ -499,6 +540,7 @@ def win_put_nonblocking(
             torch.cuda.set_device(win.device)
             ready.synchronize()
         my_p = win.associated_p
+        pending = {}  # control-plane writes, flushed in one round-trip
         for dst in sorted(dst_weights.keys()):
             w = dst_weights[dst]
             if require_mutex:
@@ -511,8 +553,13 @@ def win_put_nonblocking(
                         dst, name, me, tensor if w == 1.0 else tensor.mul(w)
                     )
                 if _ops_with_associated_p:
-                    _p_slot_set(c.store, name, dst, me, my_p * w)
-                c.store.add(_ver_key(name, dst, me), 1)
+                    pending[_p_slot_key(name, dst, me)] = repr(float(my_p * w)).encode()
+                win.put_seq[dst] = win.put_seq.get(dst, 0) + 1
+                pending[_ver_put_key(name, dst, me)] = str(win.put_seq[dst]).encode()
+                if require_mutex:
+                    # version/p must be visible before the mutex is released
+                    c.store.multi_set(pending)
+                    pending = {}
             finally:
                 if require_mutex:
                     c.store.mutex_release(_mutex_name(name, dst))
@@ -524,7 +571,8 @@ def win_put_nonblocking(
                     torch.cuda.current_stream().synchronize()
         if _ops_with_associated_p:
             win.associated_p = my_p * self_weight
-            _p_self_set(c.store, name, me, win.associated_p)
+            pending[_p_self_key(name, me)] = repr(float(win.associated_p)).encode()
+        c.store.multi_set(pending)  # one round-trip for ALL destinations
         return True
 
     return _submit_win_op(name, "win.put", job)
@@ -570,6 +618,7 @@ def win_accumulate_nonblocking(
             torch.cuda.set_device(win.device)
             ready.synchronize()
         my_p = win.associated_p
+        pending = {}
         for dst in sorted(dst_weights.keys()):
             w = dst_weights[dst]
             if require_mutex:
@@ -582,9 +631,17 @@ def win_accumulate_nonblocking(
                         dst, name, me, tensor if w == 1.0 else tensor.mul(w)
                     )
                 if _ops_with_associated_p:
+                    # read-modify-write: p slots accumulate, so the read
+                    # stays per-destination (we are the only writer)
                     old = _p_slot_get(c.store, name, dst, me)
-                    _p_slot_set(c.store, name, dst, me, old + my_p * w)
-                c.store.add(_ver_key(name, dst, me), 1)
+                    pending[_p_slot_key(name, dst, me)] = repr(
+                        float(old + my_p * w)
+                    ).encode()
+                win.put_seq[dst] = win.put_seq.get(dst, 0) + 1
+                pending[_ver_put_key(name, dst, me)] = str(win.put_seq[dst]).encode()
+                if require_mutex:
+                    c.store.multi_set(pending)
+                    pending = {}
             finally:
                 if require_mutex:
                     c.store.mutex_release(_mutex_name(name, dst))
@@ -595,7 +652,8 @@ def win_accumulate_nonblocking(
                     torch.cuda.current_stream().synchronize()
         if _ops_with_associated_p:
             win.associated_p = my_p * self_weight
-            _p_self_set(c.store, name, me, win.associated_p)
+            pending[_p_self_key(name, me)] = repr(float(win.associated_p)).encode()
+        c.store.multi_set(pending)  # one round-trip for ALL destinations
         return True
 
     return _submit_win_op(name, "win.accumulate", job)
@@ -636,6 +694,7 @@ def win_get_nonblocking(
     def job():
         if win.device.type == "cuda":
             torch.cuda.set_device(win.device)
+        pending = {}
         for src in sorted(src_weights.keys()):
             w = src_weights[src]
             if require_mutex:
@@ -655,11 +714,18 @@ def win_get_nonblocking(
                             buf.mul_(w)
                 if _ops_with_associated_p:
                     src_p = _p_self_get(c.store, name, src)
-                    _p_slot_set(c.store, name, me, src, src_p * w)
-                c.store.add(_ver_key(name, me, src), 1)
+                    pending[_p_slot_key(name, me, src)] = repr(
+                        float(src_p * w)
+                    ).encode()
+                win.get_seq[src] = win.get_seq.get(src, 0) + 1
+                pending[_ver_get_key(name, me, src)] = str(win.get_seq[src]).encode()
+                if require_mutex:
+                    c.store.multi_set(pending)
+                    pending = {}
             finally:
                 if require_mutex:
                     c.store.mutex_release(_mutex_name(name, src))
+        c.store.multi_set(pending)  # one round-trip for ALL sources
         if win.device.type == "cuda":
             torch.cuda.current_stream().synchronize()
         return True

@@ -85,6 +85,48 @@ def test_fused_adam_matches_plain_training():
         )
 
 
+def test_fused_atc_unused_head_still_steps():
+    """A param whose grad hook never fires (unused head) keeps its bucket's
+    pending set non-empty; synchronize() must still step+exchange the fired
+    params of that bucket instead of silently skipping them."""
+    import bluefog_amd as bf
+
+    if not bf._ctx().is_initialized():
+        bf.init()
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(13)
+            self.used = nn.Linear(16, 16, bias=False)
+            self.unused = nn.Linear(16, 16, bias=False)  # not in forward
+
+        def forward(self, x):
+            return self.used(x)
+
+    m_fused = M().cuda()
+    m_ref = copy.deepcopy(m_fused)
+    opt = bf.DistributedAdaptThenCombineOptimizer(
+        torch.optim.SGD(m_fused.parameters(), lr=0.1),
+        model=m_fused,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+    )
+    assert opt._fused == "sgd"
+    x = torch.randn(8, 16, device="cuda")
+    opt.zero_grad()
+    m_fused(x).sum().backward()
+    opt.step()
+    torch.cuda.synchronize()
+    # reference step on the used layer only
+    m_ref(x).sum().backward()
+    with torch.no_grad():
+        m_ref.used.weight -= 0.1 * m_ref.used.weight.grad
+    assert torch.allclose(m_fused.used.weight, m_ref.used.weight, atol=1e-5), (
+        (m_fused.used.weight - m_ref.used.weight).abs().max().item()
+    )
+    assert torch.equal(m_fused.unused.weight, m_ref.unused.weight)
+
+
 @pytest.mark.parametrize("kind", ["sgd", "adam"])
 def test_fused_atc_matches_plain_training(kind):
     """ATC at world size 1: the fused bucket kernels apply the optimizer

@@ -328,6 +328,40 @@ def w_broadcast_state():
     assert sd["state"], "optimizer state missing after broadcast"
 
 
+def w_unused_head_flush():
+    """A param whose hook never fires (unused head) must not stop the rest
+    of its bucket from being allreduced: after synchronize, fired grads are
+    exactly the cross-rank average."""
+    import bluefog_amd as bf
+
+    bf.init()
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(5)
+            self.used = nn.Linear(4, 4, bias=False)
+            self.unused = nn.Linear(4, 4, bias=False)  # not in forward
+
+        def forward(self, x):
+            return self.used(x)
+
+    model = M()
+    opt = bf.DistributedGradientAllreduceOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.1), model=model
+    )
+    x = torch.full((2, 4), float(bf.rank() + 1))
+    model(x).sum().backward()
+    opt.synchronize()
+    # dL/dW_ij = sum_b x_bj = 2*(rank+1); averaged over ranks -> size+1
+    expected = torch.full((4, 4), float(bf.size() + 1))
+    assert torch.allclose(model.used.weight.grad, expected), (
+        model.used.weight.grad
+    )
+    with opt.skip_synchronize():
+        opt.step()
+
+
 def w_duplicated_module():
     """A module used twice in forward must not double-fire communication
     (reference: duplicated-module tests)."""
@@ -422,3 +456,7 @@ def test_broadcast_state():
 
 def test_duplicated_module():
     run_dist(w_duplicated_module, 2, timeout=300)
+
+
+def test_unused_head_flush():
+    run_dist(w_unused_head_flush, 2, timeout=300)
