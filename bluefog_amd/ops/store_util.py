@@ -19,6 +19,7 @@ the xGMI data plane.
 
 
 import time
+from collections import Counter
 from datetime import timedelta
 from typing import List, Optional
 
@@ -36,15 +37,20 @@ class ControlStore:
         self._rank = rank
         self._size = size
         self._prefix = prefix
+        # round-trip accounting (one increment = one store RPC), for the
+        # host-overhead soak tests and profiles
+        self.rpc_counts = Counter()
 
     # -- raw kv ------------------------------------------------------------
     def _key(self, key: str) -> str:
         return f"{self._prefix}/{key}"
 
     def set(self, key: str, value: bytes) -> None:
+        self.rpc_counts["set"] += 1
         self._store.set(self._key(key), value)
 
     def get(self, key: str) -> bytes:
+        self.rpc_counts["get"] += 1
         return self._store.get(self._key(key))
 
     def wait(self, keys: List[str], timeout_s: Optional[float] = None) -> None:
@@ -55,12 +61,14 @@ class ControlStore:
             self._store.wait(full, timedelta(seconds=timeout_s))
 
     def add(self, key: str, amount: int) -> int:
+        self.rpc_counts["add"] += 1
         return self._store.add(self._key(key), amount)
 
     def multi_set(self, pairs: dict) -> None:
         """Write many keys in ONE store round-trip (values: bytes)."""
         if not pairs:
             return
+        self.rpc_counts["multi_set"] += 1
         self._store.multi_set([self._key(k) for k in pairs], list(pairs.values()))
 
     def multi_get(self, keys: List[str]) -> List[bytes]:
@@ -68,9 +76,11 @@ class ControlStore:
         exist (TCPStore blocks on missing keys)."""
         if not keys:
             return []
+        self.rpc_counts["multi_get"] += 1
         return self._store.multi_get([self._key(k) for k in keys])
 
     def check(self, keys: List[str]) -> bool:
+        self.rpc_counts["check"] += 1
         return self._store.check([self._key(k) for k in keys])
 
     def delete(self, key: str) -> bool:

@@ -57,6 +57,65 @@ def test_bench_default_metric_is_baseline_config():
     assert args.dtype == "fp32"
 
 
+def test_bench_self_spawn_8rank():
+    """The driver's SCALE invocation is `python bench.py --gpus N` with NO
+    rendezvous env: bench.py must self-spawn N ranks and report n_gpus=N
+    (a 1-rank number labeled n_gpus=8 would invalidate the record)."""
+    env = dict(os.environ)
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT"):
+        env.pop(k, None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(_ROOT, "bench.py"), "--gpus", "8",
+         "--model", "resnet18", "--batch-size", "1", "--steps", "2",
+         "--warmup", "1", "--device", "cpu"],
+        cwd=_ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 8, rec
+    assert rec["config"]["global_batch"] == 8, rec
+
+
+def test_bench_world_size_mismatch_fails():
+    """--gpus disagreeing with the launched world must fail loudly, not
+    report a mislabeled measurement."""
+    env = dict(os.environ)
+    env.update(RANK="0", LOCAL_RANK="0", WORLD_SIZE="1",
+               MASTER_ADDR="127.0.0.1", MASTER_PORT="29399")
+    out = subprocess.run(
+        [sys.executable, os.path.join(_ROOT, "bench.py"), "--gpus", "4",
+         "--model", "resnet18", "--batch-size", "1", "--steps", "1",
+         "--warmup", "0", "--device", "cpu"],
+        cwd=_ROOT, env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode != 0
+    assert "world size" in (out.stderr + out.stdout)
+
+
+def test_bench_config_presets():
+    """--config 2..5 presets the BASELINE.json model/batch/optimizer."""
+    sys.path.insert(0, _ROOT)
+    import bench
+
+    old = sys.argv
+    try:
+        sys.argv = ["bench.py", "--config", "3"]
+        a = bench.parse_args()
+        assert (a.model, a.batch_size, a.dist_optimizer) == (
+            "resnet50", 32, "neighbor_allreduce")
+        sys.argv = ["bench.py", "--config", "5"]
+        a = bench.parse_args()
+        assert a.model == "bert_base" and a.seq_len == 512
+        assert a.dist_optimizer == "hierarchical_neighbor_allreduce"
+        sys.argv = ["bench.py", "--config", "4", "--batch-size", "8"]
+        a = bench.parse_args()  # explicit flags win over the preset
+        assert a.batch_size == 8 and a.dist_optimizer == "win_put"
+    finally:
+        sys.argv = old
+
+
 def test_bench_hierarchical_faked_machines():
     """BASELINE config 5's code path end to end: 4 ranks, 2 faked machines
     (BLUEFOG_NODES_PER_MACHINE), hierarchical_neighbor_allreduce."""
