@@ -51,6 +51,8 @@ def submit(
     device: torch.device,
     keep_alive: Sequence[torch.Tensor] = (),
     nbytes: Optional[int] = None,
+    fingerprint: str = "",
+    fp_detail: str = "",
 ) -> int:
     """Create a handle for posted works + a post-op callback and, on GPU,
     run the callback now on the side stream (stream-ordered)."""
@@ -58,6 +60,9 @@ def submit(
     from bluefog_amd.utils import metrics
 
     metrics.record_submit(name, nbytes)
+    from bluefog_amd.ops.consistency import checker
+
+    checker().record(name, nbytes, fingerprint, fp_detail)
     timeline().start_activity(name, "COMMUNICATE")
     if device.type == "cuda":
         side = ctx().side_stream()

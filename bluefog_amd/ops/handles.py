@@ -132,11 +132,58 @@ class HandleManager:
                     get_logger().warning(
                         "op %r (handle %d) has not completed for %.0f s — "
                         "one or more peer ranks likely never submitted the "
-                        "matching operation (stalled collective/p2p)",
+                        "matching operation (stalled collective/p2p)%s",
                         name,
                         hid,
                         age,
+                        self._peer_stall_report(name),
                     )
+
+    def _peer_stall_report(self, stalled_name: str) -> str:
+        """Best-effort: publish this rank's outstanding op names to the
+        control store and read what peers have published, so the warning
+        can name which ranks appear to be missing the stalled op
+        (reference analog: the missing-rank stall report,
+        operations.cc:388-433)."""
+        try:
+            from bluefog_amd.ops.context import ctx
+
+            c = ctx()
+            if not c.is_initialized() or c.size() == 1:
+                return ""
+            with self._lock:
+                mine = sorted({h.name for h in self._handles.values() if not h.done})
+            import json as _json
+
+            c.store.set(f"stall/{c.rank()}", _json.dumps(mine).encode())
+            peers = {}
+            for r in range(c.size()):
+                if r == c.rank():
+                    continue
+                if c.store.check([f"stall/{r}"]):
+                    peers[r] = _json.loads(c.store.get(f"stall/{r}").decode())
+            silent = [r for r in range(c.size())
+                      if r != c.rank() and r not in peers]
+            matching = [r for r, ops in peers.items() if stalled_name in ops]
+            lines = [""]
+            if matching:
+                lines.append(
+                    f"  peers also stalled on {stalled_name!r}: {matching} "
+                    "(op posted everywhere; look for an earlier divergence)"
+                )
+            for r, ops in peers.items():
+                if stalled_name not in ops:
+                    lines.append(
+                        f"  rank {r} does NOT have {stalled_name!r} outstanding "
+                        f"(its oldest: {ops[:3]}) — likely the missing rank"
+                    )
+            if silent:
+                lines.append(
+                    f"  rank(s) {silent} report no stalls (running or dead)"
+                )
+            return "\n".join(lines)
+        except Exception:  # never let diagnostics take down the watchdog
+            return ""
 
     def allocate(self, name: str) -> OpHandle:
         with self._lock:

@@ -98,6 +98,14 @@ def post_neighbor_exchange_raw(
     dst_ranks = list(dst_weights.keys())
     if dynamic_enabled and enable_topo_check:
         engine.check_src_dst_consistency(src_ranks, dst_ranks, "neighbor_allreduce")
+    from bluefog_amd.ops.consistency import checker
+
+    checker().record(
+        "neighbor.allreduce.fused",
+        tensor.numel() * tensor.element_size() * (1 + len(src_ranks)),
+        f"shape={tuple(tensor.shape)},dtype={tensor.dtype}",
+        detail=f"src={sorted(src_ranks)},dst={sorted(dst_ranks)}",
+    )
     works, gathered, keep_alive = _post_neighbor_exchange(
         tensor, src_ranks, dst_ranks, dst_weights, dst_weighting_enabled
     )
@@ -147,6 +155,8 @@ def _neighbor_allreduce_nonblocking_impl(
         tensor.device,
         keep_alive=keep_alive + [output],
         nbytes=tensor.numel() * tensor.element_size() * (1 + len(src_ranks)),
+        fingerprint=f"shape={tuple(tensor.shape)},dtype={tensor.dtype}",
+        fp_detail=f"src={sorted(src_ranks)},dst={sorted(dst_ranks)}",
     )
 
 
@@ -284,6 +294,8 @@ def neighbor_allgather_nonblocking(
         finalize,
         tensor.device,
         keep_alive=[tensor, output] + recvs,
+        fingerprint=f"shape={tuple(tensor.shape)},dtype={tensor.dtype}",
+        fp_detail=f"src={sorted(src_ranks)},dst={sorted(dst_ranks)}",
     )
 
 
