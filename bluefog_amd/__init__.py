@@ -233,6 +233,7 @@ from bluefog_amd.ops.window import (  # noqa: E402
     win_free,
     win_get,
     win_get_nonblocking,
+    win_lock,
     win_mutex,
     win_poll,
     win_put,

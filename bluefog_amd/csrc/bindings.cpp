@@ -49,6 +49,14 @@ hipError_t bf_bn_relu_bwd(void* dx, const void* x, const void* dy,
                           float* ws, float* dgamma, float* dbeta, float* coef,
                           int C, long HW, long numel, bool nhwc, int dtype,
                           hipStream_t stream);
+hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,
+                         const float* gamma, const float* beta, float* mean,
+                         float* rstd, int H, long nrows, double eps, int dtype,
+                         hipStream_t stream);
+hipError_t bf_ln_add_bwd(void* dx, const void* x, const void* r,
+                         const void* dy, const float* gamma, const float* mean,
+                         const float* rstd, float* dgamma, float* dbeta, int H,
+                         long nrows, int dtype, hipStream_t stream);
 }
 
 namespace {
@@ -262,6 +270,51 @@ void bn_relu_bwd(at::Tensor dx, at::Tensor x, at::Tensor dy, at::Tensor gamma,
             "bn_relu_bwd");
 }
 
+
+void ln_add_fwd(at::Tensor y, at::Tensor x, at::Tensor r, at::Tensor gamma,
+                at::Tensor beta, at::Tensor mean, at::Tensor rstd, double eps) {
+  TORCH_CHECK(x.is_contiguous() && r.is_contiguous() && y.is_contiguous(),
+              "ln_add_fwd needs contiguous tensors");
+  TORCH_CHECK(x.sizes() == r.sizes() && y.sizes() == x.sizes(), "shape mismatch");
+  TORCH_CHECK(x.scalar_type() != at::kDouble, "ln_add_fwd: f64 unsupported");
+  const int H = static_cast<int>(x.size(-1));
+  const long nrows = x.numel() / H;
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat && gamma.numel() == H &&
+                  beta.scalar_type() == at::kFloat && beta.numel() == H &&
+                  gamma.is_contiguous() && beta.is_contiguous(),
+              "ln_add_fwd: gamma/beta must be fp32[H]");
+  TORCH_CHECK(mean.scalar_type() == at::kFloat && mean.numel() >= nrows &&
+                  rstd.scalar_type() == at::kFloat && rstd.numel() >= nrows,
+              "ln_add_fwd: mean/rstd must be fp32[rows]");
+  check_hip(bf_ln_add_fwd(y.data_ptr(), x.data_ptr(), r.data_ptr(),
+                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                          mean.data_ptr<float>(), rstd.data_ptr<float>(), H,
+                          nrows, eps, dtype_code(x), current_stream()),
+            "ln_add_fwd");
+}
+
+void ln_add_bwd(at::Tensor dx, at::Tensor x, at::Tensor r, at::Tensor dy,
+                at::Tensor gamma, at::Tensor mean, at::Tensor rstd,
+                at::Tensor dgamma, at::Tensor dbeta) {
+  TORCH_CHECK(x.is_contiguous() && r.is_contiguous() && dy.is_contiguous() &&
+                  dx.is_contiguous(),
+              "ln_add_bwd needs contiguous tensors");
+  TORCH_CHECK(x.sizes() == r.sizes() && dy.sizes() == x.sizes() &&
+                  dx.sizes() == x.sizes(),
+              "shape mismatch");
+  const int H = static_cast<int>(x.size(-1));
+  const long nrows = x.numel() / H;
+  TORCH_CHECK(dgamma.scalar_type() == at::kFloat && dgamma.numel() == H &&
+                  dbeta.scalar_type() == at::kFloat && dbeta.numel() == H,
+              "ln_add_bwd: dgamma/dbeta must be fp32[H] (zero-initialized)");
+  check_hip(bf_ln_add_bwd(dx.data_ptr(), x.data_ptr(), r.data_ptr(),
+                          dy.data_ptr(), gamma.data_ptr<float>(),
+                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                          dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), H,
+                          nrows, dtype_code(x), current_stream()),
+            "ln_add_bwd");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -281,4 +334,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused training BatchNorm2d+ReLU forward (stats + normalize)");
   m.def("bn_relu_bwd", &bn_relu_bwd,
         "fused BatchNorm2d+ReLU backward (reduce + dx, mask from x)");
+  m.def("ln_add_fwd", &ln_add_fwd,
+        "y = LayerNorm(x + r); saves per-row mean/rstd");
+  m.def("ln_add_bwd", &ln_add_bwd,
+        "dx (shared by both residual branches) + dgamma/dbeta");
 }

@@ -1048,6 +1048,187 @@ struct BnBwdLauncher {
   }
 };
 
+// ---------------------------------------------------------------------------
+// fused residual add + LayerNorm (BERT hot path): y = LN(x + r) over the
+// last dimension H. One workgroup (4 waves) per row, grid-strided over
+// rows; the row stays in REGISTERS between the stats pass and the
+// normalize pass (H <= kThreads*kLnMaxIt), so forward is 2 reads + 1
+// write per element where torch's add + native_layer_norm is 3 reads +
+// 2 writes. Backward fuses the residual branch for free (one dx feeds
+// both) and accumulates dgamma/dbeta per thread across its rows —
+// column ownership is identical for every row, so ONE atomic per owned
+// column per workgroup publishes the partials.
+// ---------------------------------------------------------------------------
+
+constexpr int kLnMaxIt = 16;  // supports H <= kThreads * kLnMaxIt = 4096
+
+// two-value workgroup tree reduction through LDS; result broadcast to all
+__device__ __forceinline__ float2 ln_block_reduce(float a, float b,
+                                                  float* lds /* [2*kThreads] */) {
+  lds[threadIdx.x] = a;
+  lds[kThreads + threadIdx.x] = b;
+  __syncthreads();
+  for (int s = kThreads / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) {
+      lds[threadIdx.x] += lds[threadIdx.x + s];
+      lds[kThreads + threadIdx.x] += lds[kThreads + threadIdx.x + s];
+    }
+    __syncthreads();
+  }
+  float2 out = make_float2(lds[0], lds[kThreads]);
+  __syncthreads();
+  return out;
+}
+
+template <typename T, int NIT>
+__global__ __launch_bounds__(kThreads) void ln_add_fwd_k(
+    T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ r,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out, int H,
+    long nrows, float eps) {
+  __shared__ float lds[2 * kThreads];
+  float v[NIT];
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const long base = row * H;
+    float s = 0.f, sq = 0.f;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int i = it * kThreads + threadIdx.x;
+      float val = 0.f;
+      if (i < H) val = AccOf<T>::to(x[base + i]) + AccOf<T>::to(r[base + i]);
+      v[it] = val;
+      s += val;
+      sq += val * val;
+    }
+    const float2 tot = ln_block_reduce(s, sq, lds);
+    const float mean = tot.x / H;
+    const float var = fmaxf(tot.y / H - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int i = it * kThreads + threadIdx.x;
+      if (i < H)
+        y[base + i] =
+            AccOf<T>::from((v[it] - mean) * rstd * gamma[i] + beta[i]);
+    }
+  }
+}
+
+template <typename T, int NIT>
+__global__ __launch_bounds__(kThreads) void ln_add_bwd_k(
+    T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ r,
+    const T* __restrict__ dy, const float* __restrict__ gamma,
+    const float* __restrict__ mean_s, const float* __restrict__ rstd_s,
+    float* __restrict__ dgamma, float* __restrict__ dbeta, int H,
+    long nrows) {
+  __shared__ float lds[2 * kThreads];
+  float dg[NIT], db[NIT];
+#pragma unroll
+  for (int it = 0; it < NIT; ++it) dg[it] = db[it] = 0.f;
+  for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const long base = row * H;
+    const float mean = mean_s[row], rstd = rstd_s[row];
+    float xh[NIT], g[NIT];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int i = it * kThreads + threadIdx.x;
+      float xhat = 0.f, gg = 0.f;
+      if (i < H) {
+        const float v = AccOf<T>::to(x[base + i]) + AccOf<T>::to(r[base + i]);
+        xhat = (v - mean) * rstd;
+        const float dyv = AccOf<T>::to(dy[base + i]);
+        gg = dyv * gamma[i];
+        dg[it] += dyv * xhat;
+        db[it] += dyv;
+      }
+      xh[it] = xhat;
+      g[it] = gg;
+      s1 += gg;
+      s2 += gg * xhat;
+    }
+    const float2 tot = ln_block_reduce(s1, s2, lds);
+    const float a1 = tot.x / H, a2 = tot.y / H;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int i = it * kThreads + threadIdx.x;
+      if (i < H)
+        dx[base + i] = AccOf<T>::from(rstd * (g[it] - a1 - xh[it] * a2));
+    }
+  }
+#pragma unroll
+  for (int it = 0; it < NIT; ++it) {
+    const int i = it * kThreads + threadIdx.x;
+    if (i < H) {
+      if (dg[it] != 0.f) atomicAdd(&dgamma[i], dg[it]);
+      if (db[it] != 0.f) atomicAdd(&dbeta[i], db[it]);
+    }
+  }
+}
+
+// NIT instantiated on {1,2,4,8,16} — the guarded extra iteration of a
+// rounded-up NIT is branch-predicated and costs nothing measurable
+#define BF_LN_NIT_DISPATCH(LAUNCH)                                            \
+  do {                                                                        \
+    const int nit = (H + kThreads - 1) / kThreads;                            \
+    if (nit > kLnMaxIt) return hipErrorInvalidValue;                          \
+    if (nit <= 1) {                                                           \
+      LAUNCH(1);                                                              \
+    } else if (nit <= 2) {                                                    \
+      LAUNCH(2);                                                              \
+    } else if (nit <= 4) {                                                    \
+      LAUNCH(4);                                                              \
+    } else if (nit <= 8) {                                                    \
+      LAUNCH(8);                                                              \
+    } else {                                                                  \
+      LAUNCH(16);                                                             \
+    }                                                                         \
+  } while (0)
+
+template <typename T>
+struct LnAddFwdLauncher {
+  static hipError_t run(void* y, const void* x, const void* r,
+                        const float* gamma, const float* beta, float* mean,
+                        float* rstd, int H, long nrows, double eps,
+                        hipStream_t stream) {
+    const int grid = static_cast<int>(nrows < 8192 ? (nrows > 0 ? nrows : 1)
+                                                   : 8192);
+#define BF_LAUNCH_LN_FWD(NIT)                                                 \
+  hipLaunchKernelGGL((ln_add_fwd_k<T, NIT>), dim3(grid), dim3(kThreads), 0,   \
+                     stream, static_cast<T*>(y), static_cast<const T*>(x),    \
+                     static_cast<const T*>(r), gamma, beta, mean, rstd, H,    \
+                     nrows, static_cast<float>(eps))
+    BF_LN_NIT_DISPATCH(BF_LAUNCH_LN_FWD);
+#undef BF_LAUNCH_LN_FWD
+    return hipGetLastError();
+  }
+};
+
+template <typename T>
+struct LnAddBwdLauncher {
+  static hipError_t run(void* dx, const void* x, const void* r, const void* dy,
+                        const float* gamma, const float* mean,
+                        const float* rstd, float* dgamma, float* dbeta, int H,
+                        long nrows, hipStream_t stream) {
+    // smaller grid than rows: each workgroup folds several rows into its
+    // register dgamma/dbeta partials before the one-atomic publish
+    const int grid = static_cast<int>(nrows < 2048 ? (nrows > 0 ? nrows : 1)
+                                                   : 2048);
+#define BF_LAUNCH_LN_BWD(NIT)                                                 \
+  hipLaunchKernelGGL((ln_add_bwd_k<T, NIT>), dim3(grid), dim3(kThreads), 0,   \
+                     stream, static_cast<T*>(dx), static_cast<const T*>(x),   \
+                     static_cast<const T*>(r), static_cast<const T*>(dy),     \
+                     gamma, mean, rstd, dgamma, dbeta, H, nrows)
+    BF_LN_NIT_DISPATCH(BF_LAUNCH_LN_BWD);
+#undef BF_LAUNCH_LN_BWD
+    return hipGetLastError();
+  }
+};
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -1128,6 +1309,23 @@ hipError_t bf_bn_relu_bwd(void* dx, const void* x, const void* dy,
   return dispatch_dtype<BnBwdLauncher>(dtype, dx, x, dy, gamma, beta, save_mean,
                                        save_invstd, ws, dgamma, dbeta, coef, C,
                                        HW, numel, nhwc, stream);
+}
+
+hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,
+                         const float* gamma, const float* beta, float* mean,
+                         float* rstd, int H, long nrows, double eps, int dtype,
+                         hipStream_t stream) {
+  return dispatch_dtype<LnAddFwdLauncher>(dtype, y, x, r, gamma, beta, mean,
+                                          rstd, H, nrows, eps, stream);
+}
+
+hipError_t bf_ln_add_bwd(void* dx, const void* x, const void* r,
+                         const void* dy, const float* gamma, const float* mean,
+                         const float* rstd, float* dgamma, float* dbeta, int H,
+                         long nrows, int dtype, hipStream_t stream) {
+  return dispatch_dtype<LnAddBwdLauncher>(dtype, dx, x, r, dy, gamma, mean,
+                                          rstd, dgamma, dbeta, H, nrows,
+                                          stream);
 }
 
 }  // extern "C"

@@ -81,19 +81,22 @@ class BertSelfAttention(nn.Module):
 class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
+        from bluefog_amd.ops.fused_modules import FusedAddLayerNorm
+
         self.attn = BertSelfAttention(cfg)
-        self.norm1 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        # residual join + LayerNorm as one gfx950 kernel (eager off-GPU)
+        self.norm1 = FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
         self.mlp = nn.Sequential(
             nn.Linear(cfg.hidden_size, cfg.intermediate_size),
             nn.GELU(),
             nn.Linear(cfg.intermediate_size, cfg.hidden_size),
         )
-        self.norm2 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.norm2 = FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
         self.dropout = nn.Dropout(cfg.dropout)
 
     def forward(self, x):
-        x = self.norm1(x + self.dropout(self.attn(x)))
-        return self.norm2(x + self.dropout(self.mlp(x)))
+        x = self.norm1(self.dropout(self.attn(x)), x)
+        return self.norm2(self.dropout(self.mlp(x)), x)
 
 
 class BertForMaskedLM(nn.Module):

@@ -232,12 +232,12 @@ class BlueFogContext:
 
     def machine_rank(self) -> int:
         self._require_init()
-        assert self.is_homogeneous(), "Only supports homogeneous environment now"
+        assert self.is_homogeneous(), "machine faking requires every machine to hold the same number of ranks"
         return self._machine_rank
 
     def machine_size(self) -> int:
         self._require_init()
-        assert self.is_homogeneous(), "Only supports homogeneous environment now"
+        assert self.is_homogeneous(), "machine faking requires every machine to hold the same number of ranks"
         return self._machine_size
 
     def is_homogeneous(self) -> bool:
@@ -322,7 +322,7 @@ class BlueFogContext:
             raise TypeError(
                 "machine topology must have the same number of nodes as bf.machine_size()."
             )
-        assert self.is_homogeneous(), "Only supports homogeneous environment now"
+        assert self.is_homogeneous(), "machine faking requires every machine to hold the same number of ranks"
         if topology_util.IsTopologyEquivalent(topology, self._machine_topology):
             logger.debug("Machine topology to set is the same as old one. Skip.")
             return True

@@ -63,17 +63,26 @@ def submit(
     from bluefog_amd.ops.consistency import checker
 
     checker().record(name, nbytes, fingerprint, fp_detail)
-    timeline().start_activity(name, "COMMUNICATE")
+    tl = timeline()
+    tl.start_activity(name, "COMMUNICATE")
     if device.type == "cuda":
         side = ctx().side_stream()
         cur = torch.cuda.current_stream()
         side.wait_stream(cur)
         with torch.cuda.stream(side):
+            if tl.enabled:
+                # hipEvent pair brackets the GPU execution window (comm
+                # wait + post-kernel); the timeline poller stamps it when
+                # it retires — execution time, not posting time
+                sev = torch.cuda.Event(enable_timing=True)
+                sev.record(side)
             for w in works:
                 w.wait()  # stream-level dependency only; host does not block
             result = finalize()
-            ev = torch.cuda.Event()
+            ev = torch.cuda.Event(enable_timing=tl.enabled)
             ev.record(side)
+        if tl.enabled:
+            tl.gpu_span(name, "COMMUNICATE_GPU", device, sev, ev)
         for t in keep_alive:
             if t is not None and t.is_cuda:
                 t.record_stream(side)
@@ -173,11 +182,11 @@ def resolve_recv_weights(
         dst_weighting_enabled = False
     else:
         if len(set(dst_weights)) != len(dst_weights):
-            raise ValueError("Argument dst_weights should only contain the unique ranks.")
+            raise ValueError("dst_weights must not list the same rank twice.")
         if self_weight is None or src_weights is None:
             raise ValueError(
-                "Arguments self_weight and src_weights should be presented if "
-                "enabling dynamic topology."
+                "Dynamic topology (dst_weights given) also requires self_weight "
+                "and src_weights."
             )
         dynamic_enabled = True
         if isinstance(dst_weights, (list, tuple)):
@@ -205,11 +214,10 @@ def resolve_recv_weights(
     elif self_weight is not None and src_weights is not None:
         if not isinstance(src_weights, dict):
             raise ValueError(
-                "Argument src_weights has to be a dictionary map from the "
-                "(in-)neighbor rank to the weights."
+                "src_weights must be a dict mapping in-neighbor rank -> weight."
             )
         if not isinstance(self_weight, float):
-            raise ValueError("Argument self_weight has to be a float for self rank.")
+            raise ValueError("self_weight must be a float.")
         if c.rank() in src_weights:
             raise ValueError(
                 "The key of src_weights should only contain other ranks "
@@ -219,11 +227,11 @@ def resolve_recv_weights(
             set(c.in_neighbor_ranks())
         ):
             raise ValueError(
-                "The key of weights should only contain the ranks that belong to "
-                " in-neighbors and self rank."
+                "src_weights may only name static in-neighbor ranks; use dynamic "
+                "mode (pass dst_weights) for arbitrary peers."
             )
     else:
         raise ValueError(
-            "Arguments self_weight and src_weights have to be presented at the same time"
+            "self_weight and src_weights must be given together (or both omitted)"
         )
     return self_weight, dict(src_weights), dst_weights_d, dynamic_enabled, dst_weighting_enabled

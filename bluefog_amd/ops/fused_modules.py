@@ -135,6 +135,73 @@ class _BNReLU(torch.autograd.Function):
         return dx, dgamma, dbeta, None, None, None, None, None
 
 
+class _AddLayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, r, weight, bias, eps):
+        from bluefog_amd import _C
+
+        x = x.contiguous()
+        r = r.contiguous()
+        H = x.shape[-1]
+        nrows = x.numel() // H
+        y = torch.empty_like(x)
+        f32 = dict(device=x.device, dtype=torch.float32)
+        mean = torch.empty(nrows, **f32)
+        rstd = torch.empty(nrows, **f32)
+        w32 = weight.contiguous()
+        b32 = bias.contiguous()
+        _C.ln_add_fwd(y, x, r, w32, b32, mean, rstd, float(eps))
+        ctx.save_for_backward(x, r, w32, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        from bluefog_amd import _C
+
+        x, r, w32, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        H = x.shape[-1]
+        dx = torch.empty_like(x)
+        dgamma = torch.zeros(H, dtype=torch.float32, device=x.device)
+        dbeta = torch.zeros_like(dgamma)
+        _C.ln_add_bwd(dx, x, r, dy, w32, mean, rstd, dgamma, dbeta)
+        # the residual join is linear: both branches share dx
+        return dx, dx, dgamma, dbeta, None
+
+
+def _ln_fusable(x: torch.Tensor, r: torch.Tensor, weight) -> bool:
+    return (
+        x.is_cuda
+        and hip_ext.has_extension()
+        and os.environ.get("BLUEFOG_FUSED_LN", "1") not in ("0", "false")
+        and x.dtype in (torch.float32, torch.float16, torch.bfloat16)
+        and r.dtype == x.dtype
+        and r.shape == x.shape
+        and weight is not None
+        and weight.dtype == torch.float32
+        and x.shape[-1] <= 4096  # kernel keeps the row in registers
+    )
+
+
+class FusedAddLayerNorm(torch.nn.LayerNorm):
+    """``forward(x, residual)`` computes ``LayerNorm(x + residual)`` as one
+    gfx950 kernel (the transformer-block join: 2 reads + 1 write per
+    element vs torch's separate add + native_layer_norm at 3 reads +
+    2 writes), with a backward that feeds both residual branches from a
+    single fused kernel. State-dict compatible with ``nn.LayerNorm``;
+    eager fallback off-GPU / for unsupported shapes (CPU tests compare
+    the paths). Reference analog: none — the reference leaves all model
+    math to torch; this follows the add_relu pattern above
+    (VERDICT r1 item 8)."""
+
+    def forward(self, x, residual=None):
+        if residual is None:
+            return super().forward(x)
+        if len(self.normalized_shape) == 1 and _ln_fusable(x, residual, self.weight):
+            return _AddLayerNorm.apply(x, residual, self.weight, self.bias, self.eps)
+        return super().forward(x + residual)
+
+
 class FusedBNReLU2d(torch.nn.BatchNorm2d):
     """BatchNorm2d immediately followed by ReLU, fused into gfx950 kernels
     in training on GPU: stats (one read of x) -> normalize+ReLU (one

@@ -180,7 +180,7 @@ def neighbor_allreduce(
     """
     if (self_weight is None) != (src_weights is None):
         raise ValueError(
-            "Arguments self_weight and src_weights have to be presented at the same time"
+            "self_weight and src_weights must be given together (or both omitted)"
         )
     handle = neighbor_allreduce_nonblocking(
         tensor,
@@ -204,7 +204,7 @@ def neighbor_allreduce_nonblocking(
 ) -> int:
     if (self_weight is None) != (src_weights is None):
         raise ValueError(
-            "Arguments self_weight and src_weights have to be presented at the same time"
+            "self_weight and src_weights must be given together (or both omitted)"
         )
     return _neighbor_allreduce_nonblocking_impl(
         tensor, self_weight, src_weights, dst_weights, enable_topo_check, name
@@ -249,7 +249,7 @@ def neighbor_allgather_nonblocking(
     c = ctx()
     if (src_ranks is None) != (dst_ranks is None):
         raise ValueError(
-            "Argument dst_ranks and src_ranks have to be presented at the same time"
+            "src_ranks and dst_ranks must be given together (or both omitted)"
         )
     dynamic = src_ranks is not None
     if dynamic and (c.rank() in src_ranks or c.rank() in dst_ranks):
@@ -380,14 +380,14 @@ def hierarchical_neighbor_allreduce_nonblocking(
     elif self_weight is not None and src_machine_weights is not None and dst_machine_weights is not None:
         if not isinstance(src_machine_weights, dict):
             raise ValueError(
-                "Argument src_machine_weights has to be a dictionary map from the "
+                "src_machine_weights must be a dict mapping machine id -> weight "
                 "(in-)neighbor rank to the weights."
             )
         if not isinstance(self_weight, float):
-            raise ValueError("Argument self_weight has to be a float for self rank.")
+            raise ValueError("self_weight must be a float.")
         if len(set(dst_machine_weights)) != len(dst_machine_weights):
             raise ValueError(
-                "Argument dst_machine_weights should only contain the unique ranks."
+                "dst_machine_weights must not list the same machine twice."
             )
         if isinstance(dst_machine_weights, (list, tuple)):
             dst_machine_weights = {int(d): 1.0 for d in dst_machine_weights}
@@ -400,7 +400,7 @@ def hierarchical_neighbor_allreduce_nonblocking(
     else:
         raise ValueError(
             "Arguments self_weight, src_machine_weights, dst_machine_weights have "
-            "to be presented at the same time."
+            "to be given together."
         )
 
     if enable_topo_check:
@@ -508,7 +508,7 @@ def pair_gossip_nonblocking(
         raise ValueError("pair_gossip target_rank must be another rank.")
     if (self_weight is None) != (pair_weight is None):
         raise ValueError(
-            "Arguments self_weight and pair_weight have to be presented at the same time"
+            "self_weight and pair_weight must be given together (or both omitted)"
         )
     if self_weight is None:
         self_weight, pair_weight = 0.5, 0.5

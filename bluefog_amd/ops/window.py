@@ -55,6 +55,7 @@ __all__ = [
     "win_poll",
     "win_wait",
     "win_mutex",
+    "win_lock",
     "get_win_version",
     "get_current_created_window_names",
     "win_associated_p",
@@ -359,6 +360,21 @@ def win_associated_p(name: str) -> float:
 
 
 @contextmanager
+def win_lock(name: str):
+    """Access-epoch context for this rank's window (reference analog:
+    win_lock, mpi_ops.py:1415-1442 — an MPI passive-target epoch over the
+    neighbor windows, mpi_controller.cc:1193-1236). This framework has no
+    RMA epochs — one-sided traffic is ordered through the distributed
+    window mutex instead — so the epoch maps to holding this rank's own
+    mutex: mutex-honoring writers (win_put/win_accumulate/win_get with
+    ``require_mutex=True``) are excluded for the duration. As in the
+    reference, plain win ops do not need this context."""
+    registry().get(name)  # raise early on unknown window
+    with win_mutex(name, for_self=True):
+        yield
+
+
+@contextmanager
 def win_mutex(name: str, for_self: bool = False, ranks: Optional[List[int]] = None):
     """Acquire the distributed window mutex of the given ranks (default: all
     out-neighbors), or of self with ``for_self=True``."""
@@ -399,11 +415,11 @@ def win_update(
     if neighbor_weights is not None and self_weight is not None:
         if not isinstance(neighbor_weights, dict):
             raise ValueError(
-                "Argument neighbor_weights has to be a dictionary map from the "
+                "neighbor_weights must be a dict mapping in-neighbor rank -> weight "
                 "(in-)neighbor rank to the weights."
             )
         if not isinstance(self_weight, float):
-            raise ValueError("Argument self_weight has to be a float for self rank.")
+            raise ValueError("self_weight must be a float.")
         if not set(neighbor_weights.keys()).issubset(set(c.in_neighbor_ranks())):
             raise ValueError(
                 "The key of weights should only contain the ranks that belong to "

@@ -29,6 +29,42 @@ from bluefog_amd.utils.logging import get_logger
 logger = get_logger()
 
 
+
+def _attach_timeline_hooks(opt):
+    """Model-level FORWARD/BACKWARD host spans (reference analog:
+    _register_timeline, optimizers.py:112-163). Together with the GPU comm
+    spans emitted by engine.submit, a trace shows whether communication
+    overlaps backward."""
+    from bluefog_amd.utils.timeline import timeline
+
+    handles = []
+    for model in getattr(opt, "_models", []):
+        def _pre(mod, inp):
+            timeline().start_activity("model.compute", "FORWARD")
+
+        def _post(mod, inp, out):
+            timeline().end_activity("model.compute")
+            timeline().start_activity("model.compute", "BACKWARD")
+
+        handles.append(model.register_forward_pre_hook(_pre))
+        handles.append(model.register_forward_hook(_post))
+    return handles
+
+
+def _detach_timeline_hooks(handles):
+    for h in handles:
+        try:
+            h.remove()
+        except Exception:
+            pass
+
+
+def _end_backward_span():
+    from bluefog_amd.utils.timeline import timeline
+
+    timeline().end_activity("model.compute")
+
+
 class CommunicationType(Enum):
     neighbor_allreduce = "neighbor.allreduce"
     hierarchical_neighbor_allreduce = "hierarchical.neighbor.allreduce"
@@ -272,11 +308,16 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 
     def turn_on_timeline(self):
         self._use_timeline = True
+        if not self._timeline_hook_handles:
+            self._timeline_hook_handles = _attach_timeline_hooks(self)
 
     def turn_off_timeline(self):
         self._use_timeline = False
+        _detach_timeline_hooks(self._timeline_hook_handles)
+        self._timeline_hook_handles = []
 
     def synchronize(self):
+        _end_backward_span()
         bf = _bf()
         with torch.no_grad():
             # only act when a communication round is underway (some bucket
@@ -560,11 +601,16 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
 
     def turn_on_timeline(self):
         self._use_timeline = True
+        if not self._timeline_hook_handles:
+            self._timeline_hook_handles = _attach_timeline_hooks(self)
 
     def turn_off_timeline(self):
         self._use_timeline = False
+        _detach_timeline_hooks(self._timeline_hook_handles)
+        self._timeline_hook_handles = []
 
     def synchronize(self):
+        _end_backward_span()
         bf = _bf()
         with torch.no_grad():
             for name, (b, handle) in self._handles.items():
@@ -979,11 +1025,16 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
 
     def turn_on_timeline(self):
         self._use_timeline = True
+        if not self._timeline_hook_handles:
+            self._timeline_hook_handles = _attach_timeline_hooks(self)
 
     def turn_off_timeline(self):
         self._use_timeline = False
+        _detach_timeline_hooks(self._timeline_hook_handles)
+        self._timeline_hook_handles = []
 
     def synchronize(self):
+        _end_backward_span()
         bf = _bf()
         from bluefog_amd.ops import hip_ext
 
@@ -1175,9 +1226,13 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
 
     def turn_on_timeline(self):
         self._use_timeline = True
+        if not self._timeline_hook_handles:
+            self._timeline_hook_handles = _attach_timeline_hooks(self)
 
     def turn_off_timeline(self):
         self._use_timeline = False
+        _detach_timeline_hooks(self._timeline_hook_handles)
+        self._timeline_hook_handles = []
 
     @contextmanager
     def skip_synchronize(self):
@@ -1188,6 +1243,7 @@ class _DistributedWinOptimizer(torch.optim.Optimizer):
             self._should_synchronize = True
 
     def synchronize(self):
+        _end_backward_span()
         bf = _bf()
         with torch.no_grad():
             for name, handle in self._handles.items():
@@ -1306,9 +1362,13 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
 
     def turn_on_timeline(self):
         self._use_timeline = True
+        if not self._timeline_hook_handles:
+            self._timeline_hook_handles = _attach_timeline_hooks(self)
 
     def turn_off_timeline(self):
         self._use_timeline = False
+        _detach_timeline_hooks(self._timeline_hook_handles)
+        self._timeline_hook_handles = []
 
     @contextmanager
     def skip_synchronize(self):
@@ -1319,6 +1379,7 @@ class _DistributedPushSumOptimizer(torch.optim.Optimizer):
             self._should_synchronize = True
 
     def synchronize(self):
+        _end_backward_span()
         bf = _bf()
         flats = {self._bucket_names[id(b["flat"])]: b["flat"] for b in self._buckets.buckets}
         with torch.no_grad():

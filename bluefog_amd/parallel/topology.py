@@ -256,10 +256,10 @@ def GetExp2DynamicSendRecvMachineRanks(
     """One-peer Exp2 schedule at machine granularity, for
     hierarchical_neighbor_allreduce (homogeneous placement only)."""
     assert (self_rank % local_size) == local_rank, (
-        "It should be used under homogeneous environment only."
+        "world size must be a multiple of nodes_per_machine (homogeneous machines)."
     )
     assert (world_size % local_size) == 0, (
-        "It should be used under homogeneous environment only."
+        "world size must be a multiple of nodes_per_machine (homogeneous machines)."
     )
     assert world_size > local_size, "It should be used under at least two machines case."
 
@@ -284,11 +284,12 @@ def GetInnerOuterRingDynamicSendRecvRanks(
     num_machines = world_size // local_size
     nodes_per_machine = local_size
     assert world_size % local_size == 0, (
-        "It should be used under homogeneous environment only."
+        "world size must be a multiple of nodes_per_machine (homogeneous machines)."
     )
     assert local_size > 2, (
-        "Do no support the case where nodes_per_machine is equal or less than 2. "
-        "Consider use hierarchical_neighbor_allreduce or GetDynamicOnePeerSendRecvRanks."
+        "nodes_per_machine must be at least 3 for the inner-outer schedule; "
+        "with 2 or fewer use hierarchical_neighbor_allreduce or "
+        "GetDynamicOnePeerSendRecvRanks instead."
     )
 
     index = 0
@@ -325,11 +326,12 @@ def GetInnerOuterExpo2DynamicSendRecvRanks(
     num_machines = world_size // local_size
     nodes_per_machine = local_size
     assert world_size % local_size == 0, (
-        "It should be used under homogeneous environment only."
+        "world size must be a multiple of nodes_per_machine (homogeneous machines)."
     )
     assert local_size > 2, (
-        "Do no support the case where nodes_per_machine is equal or less than 2. "
-        "Consider use hierarchical_neighbor_allreduce or GetDynamicOnePeerSendRecvRanks."
+        "nodes_per_machine must be at least 3 for the inner-outer schedule; "
+        "with 2 or fewer use hierarchical_neighbor_allreduce or "
+        "GetDynamicOnePeerSendRecvRanks instead."
     )
 
     exp_2_out_size = int(np.log2(num_machines - 1))

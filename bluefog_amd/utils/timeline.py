@@ -31,6 +31,9 @@ class Timeline:
         self._t0 = time.monotonic_ns()
         self._lock = threading.Lock()
         self._active = {}
+        self._gpu_anchors = {}  # device idx -> (anchor event, trace us)
+        self._gpu_pending = []
+        self._gpu_poller: Optional[threading.Thread] = None
 
     # ------------------------------------------------------------------
     def init(self, path: str, rank: int) -> None:
@@ -129,6 +132,74 @@ class Timeline:
         self._emit(
             {"name": activity, "ph": "i", "pid": pid, "tid": 0, "ts": self._now_us(), "s": "t"}
         )
+
+    # ------------------------------------------------------------------
+    # GPU-completion timestamps (reference analog: pooled cudaEvents +
+    # finalizer-thread timestamping, nccl_controller.cc:411-424,
+    # 2066-2093). Each comm op records a hipEvent pair on the side stream;
+    # a poller thread converts completed pairs to trace spans on tid=1
+    # ("GPU" lane), so a trace shows *execution* windows — and therefore
+    # backward/communication overlap — not host posting times.
+    # ------------------------------------------------------------------
+    def gpu_anchor(self, device) -> None:
+        """Anchor GPU event time to the trace clock (once per device).
+        Records + synchronizes one event — only called lazily on the first
+        GPU span, never on the steady-state hot path."""
+        import torch
+
+        idx = device.index if device.index is not None else torch.cuda.current_device()
+        if idx in self._gpu_anchors:
+            return
+        ev = torch.cuda.Event(enable_timing=True)
+        ev.record(torch.cuda.current_stream(idx))
+        ev.synchronize()
+        self._gpu_anchors[idx] = (ev, self._now_us())
+
+    def gpu_span(self, tensor_name: str, activity: str, device,
+                 start_ev, end_ev) -> None:
+        """Queue a recorded hipEvent pair; the poller emits the span when
+        the end event retires (no host stall)."""
+        if not self._enabled:
+            return
+        import torch
+
+        idx = device.index if device.index is not None else torch.cuda.current_device()
+        self.gpu_anchor(device)
+        with self._lock:
+            self._gpu_pending.append((tensor_name, activity, idx, start_ev, end_ev))
+            if self._gpu_poller is None:
+                self._gpu_poller = threading.Thread(
+                    target=self._poll_gpu, daemon=True, name="bf-timeline-gpu"
+                )
+                self._gpu_poller.start()
+
+    def _poll_gpu(self) -> None:
+        while self._enabled:
+            time.sleep(0.002)
+            done = []
+            with self._lock:
+                still = []
+                for item in self._gpu_pending:
+                    try:
+                        if item[4].query():
+                            done.append(item)
+                        else:
+                            still.append(item)
+                    except Exception:
+                        pass  # drop spans whose events died
+                self._gpu_pending = still
+            for name, activity, idx, sev, eev in done:
+                try:
+                    anchor_ev, anchor_us = self._gpu_anchors[idx]
+                    t0 = anchor_us + anchor_ev.elapsed_time(sev) * 1000.0
+                    t1 = anchor_us + anchor_ev.elapsed_time(eev) * 1000.0
+                    self._emit({
+                        "name": activity, "ph": "X", "pid": self._pid(name),
+                        "tid": 1, "ts": t0, "dur": max(t1 - t0, 0.0),
+                        "args": {"lane": "gpu"},
+                    })
+                except Exception:
+                    pass
 
 
 _timeline = Timeline()

@@ -313,3 +313,72 @@ def test_fused_bn_relu_eval_matches_torch(dev):
     x = torch.randn(4, 16, 8, 8, device=dev)
     torch.cuda.synchronize()
     assert torch.allclose(fused(x), torch.relu(ref(x)), atol=1e-5)
+
+
+@pytest.mark.parametrize(
+    "dtype,atol,gatol",
+    [
+        (torch.float32, 1e-5, 2e-3),
+        (torch.float16, 2e-2, 5e-2),
+        (torch.bfloat16, 5e-2, 5e-2),
+    ],
+)
+@pytest.mark.parametrize("shape", [(4, 128, 768), (2, 512, 768), (3, 7, 1000), (1, 1, 4096)])
+def test_ln_add_matches_torch(dev, dtype, atol, gatol, shape):
+    """Fused residual add + LayerNorm vs torch fp32 reference: forward,
+    dx (both branches) and dgamma/dbeta."""
+    from bluefog_amd.ops.fused_modules import _AddLayerNorm
+
+    torch.manual_seed(12)
+    H = shape[-1]
+    x = torch.randn(shape, device=dev, dtype=dtype, requires_grad=True)
+    r = torch.randn(shape, device=dev, dtype=dtype, requires_grad=True)
+    w = torch.randn(H, device=dev, dtype=torch.float32, requires_grad=True)
+    b = torch.randn(H, device=dev, dtype=torch.float32, requires_grad=True)
+    y = _AddLayerNorm.apply(x, r, w, b, 1e-12)
+    # fp32 torch reference
+    x32 = x.detach().float().requires_grad_()
+    r32 = r.detach().float().requires_grad_()
+    w32 = w.detach().clone().requires_grad_()
+    b32 = b.detach().clone().requires_grad_()
+    y_ref = torch.nn.functional.layer_norm(x32 + r32, (H,), w32, b32, 1e-12)
+    torch.cuda.synchronize()
+    assert torch.allclose(y.float(), y_ref, atol=atol), (
+        (y.float() - y_ref).abs().max().item()
+    )
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dtype))
+    y_ref.backward(g)
+    torch.cuda.synchronize()
+    for got, ref in [(x.grad, x32.grad), (r.grad, r32.grad)]:
+        assert torch.allclose(got.float(), ref, atol=atol * 4), (
+            (got.float() - ref).abs().max().item()
+        )
+    # column reductions accumulate over rows -> looser tolerance
+    assert torch.allclose(w.grad, w32.grad, atol=gatol * max(1.0, shape[0] * shape[1] / 512)), (
+        (w.grad - w32.grad).abs().max().item()
+    )
+    assert torch.allclose(b.grad, b32.grad, atol=gatol * max(1.0, shape[0] * shape[1] / 512)), (
+        (b.grad - b32.grad).abs().max().item()
+    )
+
+
+def test_fused_add_layernorm_module(dev):
+    """FusedAddLayerNorm(x, residual) == nn.LayerNorm(x + residual) in
+    training, including through a BertLayer-shaped composite."""
+    from bluefog_amd.ops.fused_modules import FusedAddLayerNorm
+
+    torch.manual_seed(3)
+    m = FusedAddLayerNorm(768).to(dev)
+    ref = torch.nn.LayerNorm(768).to(dev)
+    ref.load_state_dict(m.state_dict())
+    x = torch.randn(4, 64, 768, device=dev)
+    r = torch.randn(4, 64, 768, device=dev)
+    out = m(x, r)
+    expected = ref(x + r)
+    torch.cuda.synchronize()
+    assert torch.allclose(out, expected, atol=1e-5), (
+        (out - expected).abs().max().item()
+    )
+    # no-residual call degrades to plain LayerNorm
+    assert torch.allclose(m(x), ref(x), atol=1e-6)
