@@ -38,3 +38,65 @@ def test_bfrun_kills_stragglers_on_failure():
     )
     out = _run(2, code, timeout=90)
     assert out.returncode == 3
+
+
+def test_parse_hosts_and_blocks():
+    from bluefog_amd.run.bfrun import parse_hosts, rank_blocks
+
+    hosts = parse_hosts("nodeA:8, nodeB:8", 16)
+    assert hosts == [("nodeA", 8), ("nodeB", 8)]
+    assert rank_blocks(hosts, 16) == [("nodeA", 0, 8), ("nodeB", 8, 8)]
+    # -np smaller than total slots: trailing hosts get fewer/no ranks
+    assert rank_blocks(hosts, 10) == [("nodeA", 0, 8), ("nodeB", 8, 2)]
+    assert rank_blocks(hosts, 8) == [("nodeA", 0, 8)]
+    import pytest
+
+    with pytest.raises(ValueError):
+        parse_hosts("nodeA:4", 8)  # not enough slots
+    with pytest.raises(ValueError):
+        parse_hosts("nodeA:x", 1)
+
+
+def test_remote_command_composition():
+    from bluefog_amd.run.bfrun import remote_command
+
+    cmd = remote_command("nodeB", 8, 8, 16, "nodeA", 29501,
+                         ["python", "train.py", "--lr", "0.1"], ["FOO=1"])
+    assert cmd[0] == "ssh" and "nodeB" in cmd
+    joined = cmd[-1]
+    assert "--local-first-rank 8" in joined
+    assert "--local-num 8" in joined
+    assert "--master-addr nodeA" in joined
+    assert "--extra-env FOO=1" in joined
+    assert joined.endswith("-- python train.py --lr 0.1")
+
+
+def test_multihost_rank_blocks_rendezvous():
+    """Simulate the two halves of a 2-host launch as two local bfrun
+    invocations with --local-first-rank/--local-num (exactly what the ssh
+    side runs): all ranks must rendezvous into one world."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    code = (
+        "import os, torch.distributed as dist\n"
+        "dist.init_process_group('gloo')\n"
+        "print('RZ', dist.get_rank(), dist.get_world_size(), flush=True)\n"
+        "dist.barrier()\n"
+        "dist.destroy_process_group()\n"
+    )
+    halves = []
+    for first, cnt in [(0, 1), (1, 1)]:
+        halves.append(subprocess.Popen(
+            [sys.executable, "-m", "bluefog_amd.run.bfrun", "-np", "2",
+             "--master-addr", "127.0.0.1", "--master-port", str(port),
+             "--local-first-rank", str(first), "--local-num", str(cnt),
+             "--", sys.executable, "-c", code],
+            cwd=_ROOT, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True))
+    outs = [p.communicate(timeout=120) for p in halves]
+    assert all(p.returncode == 0 for p in halves), outs
+    lines = sorted(l for o, _ in outs for l in o.splitlines() if l.startswith("RZ"))
+    assert lines == ["RZ 0 2", "RZ 1 2"], outs
