@@ -116,6 +116,34 @@ def run(numel, iters, dtype, dev):
         ms = time_kernel(lambda: _C.relu_bwd_mask(o4, g4, a4), iters)
         add("relu_bwd_mask[64x256x56x56]", ms, 3 * n4 * esz)
 
+    if dtype in (torch.float32, torch.bfloat16):
+        # fused residual add + LayerNorm vs the torch chain, BERT-base
+        # shape (bs16 x seq512 x H768)
+        xb = torch.randn(16, 512, 768, device=dev, dtype=dtype)
+        rb = torch.randn_like(xb)
+        yb = torch.empty_like(xb)
+        gln = torch.rand(768, device=dev) + 0.5
+        bln = torch.randn(768, device=dev)
+        nrows = 16 * 512
+        mln = torch.empty(nrows, device=dev)
+        sln = torch.empty(nrows, device=dev)
+        nl = xb.numel()
+        ms = time_kernel(lambda: _C.ln_add_fwd(yb, xb, rb, gln, bln, mln, sln, 1e-12), iters)
+        add("ln_add_fwd[16x512x768]", ms, 3 * nl * esz)
+        ms = time_kernel(
+            lambda: torch.nn.functional.layer_norm(xb + rb, (768,), gln.to(dtype), bln.to(dtype)),
+            iters,
+        )
+        add("torch_add_ln_fwd[16x512x768]", ms, 5 * nl * esz)
+        dyb = torch.randn_like(xb)
+        dxb = torch.empty_like(xb)
+        dgln = torch.zeros(768, device=dev)
+        dbln = torch.zeros(768, device=dev)
+        ms = time_kernel(
+            lambda: _C.ln_add_bwd(dxb, xb, rb, dyb, gln, mln, sln, dgln, dbln), iters
+        )
+        add("ln_add_bwd[16x512x768]", ms, 4 * nl * esz)
+
     if dtype == torch.float32:
         # fused BN+ReLU pipeline (opt-in path) on a ResNet50-stem-shaped
         # activation: measures how far the reductions are from peak

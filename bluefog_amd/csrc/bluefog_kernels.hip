@@ -1062,43 +1062,58 @@ struct BnBwdLauncher {
 
 constexpr int kLnMaxIt = 16;  // supports H <= kThreads * kLnMaxIt = 4096
 
-// two-value workgroup tree reduction through LDS; result broadcast to all
+// two-value workgroup reduction: 64-wide wavefront butterflies (shuffles,
+// no LDS traffic) then one cross-wave fold through 8 LDS floats — two
+// barriers total, vs ~16 for a full LDS tree
 __device__ __forceinline__ float2 ln_block_reduce(float a, float b,
-                                                  float* lds /* [2*kThreads] */) {
-  lds[threadIdx.x] = a;
-  lds[kThreads + threadIdx.x] = b;
-  __syncthreads();
-  for (int s = kThreads / 2; s > 0; s >>= 1) {
-    if (threadIdx.x < s) {
-      lds[threadIdx.x] += lds[threadIdx.x + s];
-      lds[kThreads + threadIdx.x] += lds[kThreads + threadIdx.x + s];
-    }
-    __syncthreads();
+                                                  float* lds /* [8] */) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    a += __shfl_xor(a, off, 64);
+    b += __shfl_xor(b, off, 64);
   }
-  float2 out = make_float2(lds[0], lds[kThreads]);
+  __syncthreads();  // lds free from any previous call
+  if ((threadIdx.x & 63) == 0) {
+    const int wave = threadIdx.x >> 6;
+    lds[wave] = a;
+    lds[4 + wave] = b;
+  }
   __syncthreads();
-  return out;
+  return make_float2(lds[0] + lds[1] + lds[2] + lds[3],
+                     lds[4] + lds[5] + lds[6] + lds[7]);
 }
 
-template <typename T, int NIT>
+// NIT = per-thread iterations; VEC = elements per access (16B packs when
+// H divides the pack width, else scalar). A thread's row slice lives in
+// registers between the stats pass and the normalize pass.
+template <typename T, int NIT, int VEC>
 __global__ __launch_bounds__(kThreads) void ln_add_fwd_k(
     T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ r,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ rstd_out, int H,
     long nrows, float eps) {
-  __shared__ float lds[2 * kThreads];
-  float v[NIT];
+  __shared__ float lds[8];
+  float v[NIT * VEC];
   for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
     const long base = row * H;
     float s = 0.f, sq = 0.f;
 #pragma unroll
     for (int it = 0; it < NIT; ++it) {
-      const int i = it * kThreads + threadIdx.x;
-      float val = 0.f;
-      if (i < H) val = AccOf<T>::to(x[base + i]) + AccOf<T>::to(r[base + i]);
-      v[it] = val;
-      s += val;
-      sq += val * val;
+      const int col = (it * kThreads + threadIdx.x) * VEC;
+      if (col < H) {
+        Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base + col);
+        Pack<T, VEC> rv = *reinterpret_cast<const Pack<T, VEC>*>(r + base + col);
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          const float val = AccOf<T>::to(xv.v[u]) + AccOf<T>::to(rv.v[u]);
+          v[it * VEC + u] = val;
+          s += val;
+          sq += val * val;
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) v[it * VEC + u] = 0.f;
+      }
     }
     const float2 tot = ln_block_reduce(s, sq, lds);
     const float mean = tot.x / H;
@@ -1110,82 +1125,128 @@ __global__ __launch_bounds__(kThreads) void ln_add_fwd_k(
     }
 #pragma unroll
     for (int it = 0; it < NIT; ++it) {
-      const int i = it * kThreads + threadIdx.x;
-      if (i < H)
-        y[base + i] =
-            AccOf<T>::from((v[it] - mean) * rstd * gamma[i] + beta[i]);
+      const int col = (it * kThreads + threadIdx.x) * VEC;
+      if (col < H) {
+        Pack<T, VEC> ov;
+#pragma unroll
+        for (int u = 0; u < VEC; ++u)
+          ov.v[u] = AccOf<T>::from((v[it * VEC + u] - mean) * rstd *
+                                       gamma[col + u] +
+                                   beta[col + u]);
+        *reinterpret_cast<Pack<T, VEC>*>(y + base + col) = ov;
+      }
     }
   }
 }
 
-template <typename T, int NIT>
+template <typename T, int NIT, int VEC>
 __global__ __launch_bounds__(kThreads) void ln_add_bwd_k(
     T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ r,
     const T* __restrict__ dy, const float* __restrict__ gamma,
     const float* __restrict__ mean_s, const float* __restrict__ rstd_s,
     float* __restrict__ dgamma, float* __restrict__ dbeta, int H,
     long nrows) {
-  __shared__ float lds[2 * kThreads];
-  float dg[NIT], db[NIT];
+  __shared__ float lds[8];
+  float dg[NIT * VEC], db[NIT * VEC];
 #pragma unroll
-  for (int it = 0; it < NIT; ++it) dg[it] = db[it] = 0.f;
+  for (int q = 0; q < NIT * VEC; ++q) dg[q] = db[q] = 0.f;
   for (long row = blockIdx.x; row < nrows; row += gridDim.x) {
     const long base = row * H;
     const float mean = mean_s[row], rstd = rstd_s[row];
-    float xh[NIT], g[NIT];
+    float xh[NIT * VEC], g[NIT * VEC];
     float s1 = 0.f, s2 = 0.f;
 #pragma unroll
     for (int it = 0; it < NIT; ++it) {
-      const int i = it * kThreads + threadIdx.x;
-      float xhat = 0.f, gg = 0.f;
-      if (i < H) {
-        const float v = AccOf<T>::to(x[base + i]) + AccOf<T>::to(r[base + i]);
-        xhat = (v - mean) * rstd;
-        const float dyv = AccOf<T>::to(dy[base + i]);
-        gg = dyv * gamma[i];
-        dg[it] += dyv * xhat;
-        db[it] += dyv;
+      const int col = (it * kThreads + threadIdx.x) * VEC;
+      if (col < H) {
+        Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base + col);
+        Pack<T, VEC> rv = *reinterpret_cast<const Pack<T, VEC>*>(r + base + col);
+        Pack<T, VEC> yv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base + col);
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          const int q = it * VEC + u;
+          const float val = AccOf<T>::to(xv.v[u]) + AccOf<T>::to(rv.v[u]);
+          const float xhat = (val - mean) * rstd;
+          const float dyv = AccOf<T>::to(yv.v[u]);
+          const float gg = dyv * gamma[col + u];
+          xh[q] = xhat;
+          g[q] = gg;
+          dg[q] += dyv * xhat;
+          db[q] += dyv;
+          s1 += gg;
+          s2 += gg * xhat;
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          xh[it * VEC + u] = 0.f;
+          g[it * VEC + u] = 0.f;
+        }
       }
-      xh[it] = xhat;
-      g[it] = gg;
-      s1 += gg;
-      s2 += gg * xhat;
     }
     const float2 tot = ln_block_reduce(s1, s2, lds);
     const float a1 = tot.x / H, a2 = tot.y / H;
 #pragma unroll
     for (int it = 0; it < NIT; ++it) {
-      const int i = it * kThreads + threadIdx.x;
-      if (i < H)
-        dx[base + i] = AccOf<T>::from(rstd * (g[it] - a1 - xh[it] * a2));
+      const int col = (it * kThreads + threadIdx.x) * VEC;
+      if (col < H) {
+        Pack<T, VEC> ov;
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          const int q = it * VEC + u;
+          ov.v[u] = AccOf<T>::from(rstd * (g[q] - a1 - xh[q] * a2));
+        }
+        *reinterpret_cast<Pack<T, VEC>*>(dx + base + col) = ov;
+      }
     }
   }
+  // column ownership is identical for every row -> one atomic per owned
+  // column per workgroup publishes the dgamma/dbeta partials
 #pragma unroll
   for (int it = 0; it < NIT; ++it) {
-    const int i = it * kThreads + threadIdx.x;
-    if (i < H) {
-      if (dg[it] != 0.f) atomicAdd(&dgamma[i], dg[it]);
-      if (db[it] != 0.f) atomicAdd(&dbeta[i], db[it]);
+    const int col = (it * kThreads + threadIdx.x) * VEC;
+    if (col < H) {
+#pragma unroll
+      for (int u = 0; u < VEC; ++u) {
+        const int q = it * VEC + u;
+        if (dg[q] != 0.f) atomicAdd(&dgamma[col + u], dg[q]);
+        if (db[q] != 0.f) atomicAdd(&dbeta[col + u], db[q]);
+      }
     }
   }
 }
 
-// NIT instantiated on {1,2,4,8,16} — the guarded extra iteration of a
-// rounded-up NIT is branch-predicated and costs nothing measurable
-#define BF_LN_NIT_DISPATCH(LAUNCH)                                            \
+// pick (NIT, VEC): vectorized 16B packs when every access stays in-row
+// (H % VEC == 0; torch allocations are 256B-aligned so row bases stay
+// 16B-aligned), scalar otherwise. NIT instantiated on {1,2,4} vec /
+// {1,2,4,8,16} scalar — a rounded-up NIT's guarded extra iteration is
+// branch-predicated and free.
+#define BF_LN_DISPATCH(LAUNCH)                                                \
   do {                                                                        \
-    const int nit = (H + kThreads - 1) / kThreads;                            \
-    if (nit > kLnMaxIt) return hipErrorInvalidValue;                          \
-    if (nit <= 1) {                                                           \
-      LAUNCH(1);                                                              \
-    } else if (nit <= 2) {                                                    \
-      LAUNCH(2);                                                              \
-    } else if (nit <= 4) {                                                    \
-      LAUNCH(4);                                                              \
-    } else if (nit <= 8) {                                                    \
-      LAUNCH(8);                                                              \
+    constexpr int V = vec_width<T>();                                         \
+    if (H % V == 0 && H <= kThreads * kLnMaxIt) {                             \
+      const int nit = (H + kThreads * V - 1) / (kThreads * V);                \
+      if (nit <= 1) {                                                         \
+        LAUNCH(1, V);                                                         \
+      } else if (nit <= 2) {                                                  \
+        LAUNCH(2, V);                                                         \
+      } else {                                                                \
+        LAUNCH(4, V);                                                         \
+      }                                                                       \
     } else {                                                                  \
-      LAUNCH(16);                                                             \
+      const int nit = (H + kThreads - 1) / kThreads;                          \
+      if (nit > kLnMaxIt) return hipErrorInvalidValue;                        \
+      if (nit <= 1) {                                                         \
+        LAUNCH(1, 1);                                                         \
+      } else if (nit <= 2) {                                                  \
+        LAUNCH(2, 1);                                                         \
+      } else if (nit <= 4) {                                                  \
+        LAUNCH(4, 1);                                                         \
+      } else if (nit <= 8) {                                                  \
+        LAUNCH(8, 1);                                                         \
+      } else {                                                                \
+        LAUNCH(16, 1);                                                        \
+      }                                                                       \
     }                                                                         \
   } while (0)
 
@@ -1197,12 +1258,12 @@ struct LnAddFwdLauncher {
                         hipStream_t stream) {
     const int grid = static_cast<int>(nrows < 8192 ? (nrows > 0 ? nrows : 1)
                                                    : 8192);
-#define BF_LAUNCH_LN_FWD(NIT)                                                 \
-  hipLaunchKernelGGL((ln_add_fwd_k<T, NIT>), dim3(grid), dim3(kThreads), 0,   \
-                     stream, static_cast<T*>(y), static_cast<const T*>(x),    \
+#define BF_LAUNCH_LN_FWD(NIT, VV)                                             \
+  hipLaunchKernelGGL((ln_add_fwd_k<T, NIT, VV>), dim3(grid), dim3(kThreads),  \
+                     0, stream, static_cast<T*>(y), static_cast<const T*>(x), \
                      static_cast<const T*>(r), gamma, beta, mean, rstd, H,    \
                      nrows, static_cast<float>(eps))
-    BF_LN_NIT_DISPATCH(BF_LAUNCH_LN_FWD);
+    BF_LN_DISPATCH(BF_LAUNCH_LN_FWD);
 #undef BF_LAUNCH_LN_FWD
     return hipGetLastError();
   }
@@ -1218,12 +1279,12 @@ struct LnAddBwdLauncher {
     // register dgamma/dbeta partials before the one-atomic publish
     const int grid = static_cast<int>(nrows < 2048 ? (nrows > 0 ? nrows : 1)
                                                    : 2048);
-#define BF_LAUNCH_LN_BWD(NIT)                                                 \
-  hipLaunchKernelGGL((ln_add_bwd_k<T, NIT>), dim3(grid), dim3(kThreads), 0,   \
-                     stream, static_cast<T*>(dx), static_cast<const T*>(x),   \
+#define BF_LAUNCH_LN_BWD(NIT, VV)                                             \
+  hipLaunchKernelGGL((ln_add_bwd_k<T, NIT, VV>), dim3(grid), dim3(kThreads),  \
+                     0, stream, static_cast<T*>(dx), static_cast<const T*>(x),\
                      static_cast<const T*>(r), static_cast<const T*>(dy),     \
                      gamma, mean, rstd, dgamma, dbeta, H, nrows)
-    BF_LN_NIT_DISPATCH(BF_LAUNCH_LN_BWD);
+    BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
 #undef BF_LAUNCH_LN_BWD
     return hipGetLastError();
   }

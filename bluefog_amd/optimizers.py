@@ -1043,9 +1043,11 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
                 # flush partially-fired buckets (a frozen/unused param keeps
                 # the pending set non-empty so _fused_bucket_step never ran):
                 # the fired params must still be stepped and exchanged, and
-                # the collective must run on every rank
+                # the collective must run on every rank. A fully-fired
+                # bucket leaves an EMPTY pending set behind — it already
+                # stepped, so only a truthy (non-empty) set may flush.
                 for b in self._buckets.buckets:
-                    if b["name"] in self._pending and b["name"] not in self._handles:
+                    if self._pending.get(b["name"]) and b["name"] not in self._handles:
                         self._fused_bucket_step(b)
             for key, handle in self._handles.items():
                 if isinstance(handle, tuple) and handle and handle[0] == "fused":
