@@ -523,8 +523,19 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
             )
 
     def _register_hooks(self):
+        # post the parameter exchange from a forward PRE-hook: the sends
+        # only READ p.data (forward never mutates parameters), so launching
+        # before forward widens the comm/compute overlap window from
+        # [backward] to [forward + backward] — on 8 xGMI-linked GPUs the
+        # one-peer bucket exchange then hides entirely. The reference fires
+        # from post-forward module hooks (optimizers.py:354-392);
+        # BLUEFOG_AWC_HOOK=post restores that timing.
+        pre = os.environ.get("BLUEFOG_AWC_HOOK", "pre") != "post"
         for model in self._models:
-            model.register_forward_hook(self._make_hook())
+            if pre:
+                model.register_forward_pre_hook(self._make_hook())
+            else:
+                model.register_forward_hook(self._make_hook())
 
     def _make_hook(self):
         def hook(model, *unused):
