@@ -140,7 +140,7 @@ class InteractiveClient:
         self.profile = profile
         self._socks: List[socket.socket] = []
         for port in self._state["control_ports"]:
-            s = socket.create_connection(("127.0.0.1", port), timeout=60)
+            s = socket.create_connection(("127.0.0.1", port), timeout=120)
             self._socks.append(s)
 
     @property
@@ -253,7 +253,7 @@ def start_cluster(np_: int, profile: str = "bluefog", extra_env=None) -> dict:
     with open(path, "w") as f:
         json.dump(state, f)
     # wait until every worker accepts connections
-    deadline = time.time() + 60
+    deadline = time.time() + float(os.environ.get("BLUEFOG_IBFRUN_STARTUP_TIMEOUT", "180"))
     for port in control_ports:
         while True:
             try:
@@ -261,7 +261,7 @@ def start_cluster(np_: int, profile: str = "bluefog", extra_env=None) -> dict:
                     break
             except OSError:
                 if time.time() > deadline:
-                    raise ClusterError("ibfrun workers did not come up in 60 s")
+                    raise ClusterError("ibfrun workers did not come up before the startup timeout")
                 time.sleep(0.1)
     return state
 
