@@ -181,6 +181,33 @@ def w_win_mutex():
     bf.win_free("wm")
 
 
+def w_win_lock():
+    """win_lock excludes mutex-honoring writers from this rank's window:
+    a put issued under the peer's lock must not land until release."""
+    bf = _init_ring()
+    rank = bf.rank()
+    t = torch.zeros(2)
+    bf.win_create(t, "wl")
+    bf.barrier()
+    from bluefog_amd.ops.context import ctx
+
+    store = ctx().store
+    if rank == 0:
+        with bf.win_lock("wl"):
+            store.set("test/wl/locked", b"1")
+            # peer's mutex-honoring put must block while we hold the epoch
+            time.sleep(0.3)
+            assert bf.get_win_version("wl")[1] == 0, "put landed inside the epoch"
+        store.wait(["test/wl/done"], timeout_s=60)
+        assert bf.get_win_version("wl")[1] == 1
+    elif rank == 1:
+        store.wait(["test/wl/locked"], timeout_s=60)
+        bf.win_put(torch.ones(2), "wl", dst_weights={0: 1.0}, require_mutex=True)
+        store.set("test/wl/done", b"1")
+    bf.barrier()
+    bf.win_free("wl")
+
+
 def w_associated_p():
     bf = _init_ring()
     rank, size = bf.rank(), bf.size()
@@ -271,6 +298,10 @@ def test_win_version():
 
 def test_win_mutex():
     run_dist(w_win_mutex, 4)
+
+
+def test_win_lock():
+    run_dist(w_win_lock, 2, timeout=300)
 
 
 def test_associated_p():
