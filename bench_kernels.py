@@ -139,8 +139,10 @@ def run(numel, iters, dtype, dev):
         dxb = torch.empty_like(xb)
         dgln = torch.zeros(768, device=dev)
         dbln = torch.zeros(768, device=dev)
+        scr = torch.empty(2 * 768 * _C.ln_bwd_scratch_rows(nrows), device=dev)
         ms = time_kernel(
-            lambda: _C.ln_add_bwd(dxb, xb, rb, dyb, gln, mln, sln, dgln, dbln), iters
+            lambda: _C.ln_add_bwd(dxb, xb, rb, dyb, gln, mln, sln, dgln, dbln, scr),
+            iters,
         )
         add("ln_add_bwd[16x512x768]", ms, 4 * nl * esz)
 

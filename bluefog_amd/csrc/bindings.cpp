@@ -55,8 +55,10 @@ hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,
                          hipStream_t stream);
 hipError_t bf_ln_add_bwd(void* dx, const void* x, const void* r,
                          const void* dy, const float* gamma, const float* mean,
-                         const float* rstd, float* dgamma, float* dbeta, int H,
-                         long nrows, int dtype, hipStream_t stream);
+                         const float* rstd, float* dgamma, float* dbeta,
+                         float* scratch, int H, long nrows, int dtype,
+                         hipStream_t stream);
+int bf_ln_bwd_scratch_rows(long nrows);
 }
 
 namespace {
@@ -295,7 +297,7 @@ void ln_add_fwd(at::Tensor y, at::Tensor x, at::Tensor r, at::Tensor gamma,
 
 void ln_add_bwd(at::Tensor dx, at::Tensor x, at::Tensor r, at::Tensor dy,
                 at::Tensor gamma, at::Tensor mean, at::Tensor rstd,
-                at::Tensor dgamma, at::Tensor dbeta) {
+                at::Tensor dgamma, at::Tensor dbeta, at::Tensor scratch) {
   TORCH_CHECK(x.is_contiguous() && r.is_contiguous() && dy.is_contiguous() &&
                   dx.is_contiguous(),
               "ln_add_bwd needs contiguous tensors");
@@ -307,11 +309,15 @@ void ln_add_bwd(at::Tensor dx, at::Tensor x, at::Tensor r, at::Tensor dy,
   TORCH_CHECK(dgamma.scalar_type() == at::kFloat && dgamma.numel() == H &&
                   dbeta.scalar_type() == at::kFloat && dbeta.numel() == H,
               "ln_add_bwd: dgamma/dbeta must be fp32[H] (zero-initialized)");
+  TORCH_CHECK(scratch.scalar_type() == at::kFloat &&
+                  scratch.numel() >= 2L * H * bf_ln_bwd_scratch_rows(nrows),
+              "ln_add_bwd: scratch must be fp32[2*H*ln_bwd_scratch_rows]");
   check_hip(bf_ln_add_bwd(dx.data_ptr(), x.data_ptr(), r.data_ptr(),
                           dy.data_ptr(), gamma.data_ptr<float>(),
                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                          dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), H,
-                          nrows, dtype_code(x), current_stream()),
+                          dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                          scratch.data_ptr<float>(), H, nrows, dtype_code(x),
+                          current_stream()),
             "ln_add_bwd");
 }
 
@@ -338,4 +344,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "y = LayerNorm(x + r); saves per-row mean/rstd");
   m.def("ln_add_bwd", &ln_add_bwd,
         "dx (shared by both residual branches) + dgamma/dbeta");
+  m.def("ln_bwd_scratch_rows",
+        [](long nrows) { return bf_ln_bwd_scratch_rows(nrows); },
+        "workgroup slots the ln_add_bwd scratch buffer must provide");
 }

@@ -1144,8 +1144,7 @@ __global__ __launch_bounds__(kThreads) void ln_add_bwd_k(
     T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ r,
     const T* __restrict__ dy, const float* __restrict__ gamma,
     const float* __restrict__ mean_s, const float* __restrict__ rstd_s,
-    float* __restrict__ dgamma, float* __restrict__ dbeta, int H,
-    long nrows) {
+    float* __restrict__ scratch /* [gridDim.x][2H] */, int H, long nrows) {
   __shared__ float lds[8];
   float dg[NIT * VEC], db[NIT * VEC];
 #pragma unroll
@@ -1200,8 +1199,10 @@ __global__ __launch_bounds__(kThreads) void ln_add_bwd_k(
       }
     }
   }
-  // column ownership is identical for every row -> one atomic per owned
-  // column per workgroup publishes the dgamma/dbeta partials
+  // column ownership is identical for every row, so each thread holds the
+  // full per-workgroup partial for its columns: publish with PLAIN stores
+  // into the workgroup's scratch slot (atomics to H shared addresses from
+  // thousands of workgroups serialize on HBM and were 10x slower)
 #pragma unroll
   for (int it = 0; it < NIT; ++it) {
     const int col = (it * kThreads + threadIdx.x) * VEC;
@@ -1209,11 +1210,27 @@ __global__ __launch_bounds__(kThreads) void ln_add_bwd_k(
 #pragma unroll
       for (int u = 0; u < VEC; ++u) {
         const int q = it * VEC + u;
-        if (dg[q] != 0.f) atomicAdd(&dgamma[col + u], dg[q]);
-        if (db[q] != 0.f) atomicAdd(&dbeta[col + u], db[q]);
+        scratch[static_cast<long>(blockIdx.x) * 2 * H + col + u] = dg[q];
+        scratch[static_cast<long>(blockIdx.x) * 2 * H + H + col + u] = db[q];
       }
     }
   }
+}
+
+// reduce the [G][2H] scratch partials into dgamma/dbeta (coalesced: lane
+// per column, loop over workgroup slots)
+__global__ __launch_bounds__(256) void ln_bwd_finalize_k(
+    const float* __restrict__ scratch, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int H, int G) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  float sg = 0.f, sb = 0.f;
+  for (int b = 0; b < G; ++b) {
+    sg += scratch[static_cast<long>(b) * 2 * H + col];
+    sb += scratch[static_cast<long>(b) * 2 * H + H + col];
+  }
+  dgamma[col] += sg;  // caller zero-inits: += keeps accumulate semantics
+  dbeta[col] += sb;
 }
 
 // pick (NIT, VEC): vectorized 16B packs when every access stays in-row
@@ -1269,23 +1286,31 @@ struct LnAddFwdLauncher {
   }
 };
 
+constexpr int kLnBwdMaxGrid = 1024;
+
+inline int ln_bwd_grid(long nrows) {
+  return static_cast<int>(nrows < kLnBwdMaxGrid ? (nrows > 0 ? nrows : 1)
+                                                : kLnBwdMaxGrid);
+}
+
 template <typename T>
 struct LnAddBwdLauncher {
   static hipError_t run(void* dx, const void* x, const void* r, const void* dy,
                         const float* gamma, const float* mean,
-                        const float* rstd, float* dgamma, float* dbeta, int H,
-                        long nrows, hipStream_t stream) {
-    // smaller grid than rows: each workgroup folds several rows into its
-    // register dgamma/dbeta partials before the one-atomic publish
-    const int grid = static_cast<int>(nrows < 2048 ? (nrows > 0 ? nrows : 1)
-                                                   : 2048);
+                        const float* rstd, float* dgamma, float* dbeta,
+                        float* scratch, int H, long nrows,
+                        hipStream_t stream) {
+    const int grid = ln_bwd_grid(nrows);
 #define BF_LAUNCH_LN_BWD(NIT, VV)                                             \
   hipLaunchKernelGGL((ln_add_bwd_k<T, NIT, VV>), dim3(grid), dim3(kThreads),  \
                      0, stream, static_cast<T*>(dx), static_cast<const T*>(x),\
                      static_cast<const T*>(r), static_cast<const T*>(dy),     \
-                     gamma, mean, rstd, dgamma, dbeta, H, nrows)
+                     gamma, mean, rstd, scratch, H, nrows)
     BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
 #undef BF_LAUNCH_LN_BWD
+    BF_CHECK_HIP(hipGetLastError());
+    hipLaunchKernelGGL(ln_bwd_finalize_k, dim3((H + 255) / 256), dim3(256), 0,
+                       stream, scratch, dgamma, dbeta, H, grid);
     return hipGetLastError();
   }
 };
@@ -1382,11 +1407,14 @@ hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,
 
 hipError_t bf_ln_add_bwd(void* dx, const void* x, const void* r,
                          const void* dy, const float* gamma, const float* mean,
-                         const float* rstd, float* dgamma, float* dbeta, int H,
-                         long nrows, int dtype, hipStream_t stream) {
+                         const float* rstd, float* dgamma, float* dbeta,
+                         float* scratch, int H, long nrows, int dtype,
+                         hipStream_t stream) {
   return dispatch_dtype<LnAddBwdLauncher>(dtype, dx, x, r, dy, gamma, mean,
-                                          rstd, dgamma, dbeta, H, nrows,
-                                          stream);
+                                          rstd, dgamma, dbeta, scratch, H,
+                                          nrows, stream);
 }
+
+int bf_ln_bwd_scratch_rows(long nrows) { return ln_bwd_grid(nrows); }
 
 }  // extern "C"

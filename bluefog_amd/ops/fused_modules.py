@@ -161,10 +161,13 @@ class _AddLayerNorm(torch.autograd.Function):
         x, r, w32, mean, rstd = ctx.saved_tensors
         dy = dy.contiguous()
         H = x.shape[-1]
+        nrows = x.numel() // H
         dx = torch.empty_like(x)
         dgamma = torch.zeros(H, dtype=torch.float32, device=x.device)
         dbeta = torch.zeros_like(dgamma)
-        _C.ln_add_bwd(dx, x, r, dy, w32, mean, rstd, dgamma, dbeta)
+        scratch = torch.empty(2 * H * _C.ln_bwd_scratch_rows(nrows),
+                              dtype=torch.float32, device=x.device)
+        _C.ln_add_bwd(dx, x, r, dy, w32, mean, rstd, dgamma, dbeta, scratch)
         # the residual join is linear: both branches share dx
         return dx, dx, dgamma, dbeta, None
 

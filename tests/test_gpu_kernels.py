@@ -354,11 +354,15 @@ def test_ln_add_matches_torch(dev, dtype, atol, gatol, shape):
         assert torch.allclose(got.float(), ref, atol=atol * 4), (
             (got.float() - ref).abs().max().item()
         )
-    # column reductions accumulate over rows -> looser tolerance
-    assert torch.allclose(w.grad, w32.grad, atol=gatol * max(1.0, shape[0] * shape[1] / 512)), (
+    # column reductions sum `rows` input-precision-rounded products of
+    # O(1) values: the error random-walks as sqrt(rows) (and so does the
+    # magnitude of the sum — relative precision is preserved)
+    rows = shape[0] * shape[1]
+    col_atol = gatol * max(1.0, rows ** 0.5)
+    assert torch.allclose(w.grad, w32.grad, atol=col_atol), (
         (w.grad - w32.grad).abs().max().item()
     )
-    assert torch.allclose(b.grad, b32.grad, atol=gatol * max(1.0, shape[0] * shape[1] / 512)), (
+    assert torch.allclose(b.grad, b32.grad, atol=col_atol), (
         (b.grad - b32.grad).abs().max().item()
     )
 

@@ -49,14 +49,15 @@ def w_timeline_gpu_spans():
     """GPU lane: comm ops must produce tid=1 "X" spans with hipEvent-based
     durations, and the optimizer's FORWARD/BACKWARD host spans must appear
     so overlap is readable off the trace (reference analog:
-    nccl_controller.cc:411-424 GPU-completion timestamps)."""
+    nccl_controller.cc:411-424 GPU-completion timestamps). Single rank on
+    the test GPU (RCCL forbids two ranks per device); the engine.submit
+    span path is identical at any world size."""
     import bluefog_amd as bf
     import torch.nn as nn
 
     bf.init()
     rank = bf.rank()
-    torch.cuda.set_device(0)  # both ranks share the test GPU
-    bf.set_topology(bf.RingGraph(bf.size()))
+    torch.cuda.set_device(0)
     model = nn.Linear(256, 256).cuda()
     opt = bf.DistributedAdaptWithCombineOptimizer(
         torch.optim.SGD(model.parameters(), lr=0.01),
@@ -92,5 +93,8 @@ import pytest  # noqa: E402
 
 @pytest.mark.gpu
 def test_timeline_gpu_spans():
-    run_dist(w_timeline_gpu_spans, 2,
-             env={"BLUEFOG_TIMELINE": _TL_BASE + "_gpu"}, timeout=300)
+    # FUSED_STEP=0: the non-fused AWC path routes through engine.submit,
+    # which is where the GPU spans are emitted
+    run_dist(w_timeline_gpu_spans, 1,
+             env={"BLUEFOG_TIMELINE": _TL_BASE + "_gpu",
+                  "BLUEFOG_FUSED_STEP": "0"}, timeout=300)
