@@ -20,6 +20,21 @@ import time
 from typing import Optional
 
 
+def _roctx():
+    """Optional roctx/NVTX bridge (SURVEY §5.1 bonus): BLUEFOG_ROCTX=1
+    mirrors every activity as a roctx range so rocprofv3's marker trace
+    (--marker-trace) aligns framework phases with kernel timelines.
+    torch.cuda.nvtx lowers to roctx on ROCm builds."""
+    if os.environ.get("BLUEFOG_ROCTX", "0") != "1":
+        return None
+    try:
+        import torch
+
+        return torch.cuda.nvtx
+    except Exception:
+        return None
+
+
 class Timeline:
     def __init__(self):
         self._enabled = False
@@ -32,6 +47,7 @@ class Timeline:
         self._lock = threading.Lock()
         self._active = {}
         self._gpu_anchors = {}  # device idx -> (anchor event, trace us)
+        self._nvtx = _roctx()
         self._gpu_pending = []
         self._gpu_poller: Optional[threading.Thread] = None
 
@@ -106,6 +122,8 @@ class Timeline:
 
     # ------------------------------------------------------------------
     def start_activity(self, tensor_name: str, activity: str) -> bool:
+        if self._nvtx is not None:
+            self._nvtx.range_push(f"{tensor_name}:{activity}")
         if not self._enabled:
             return False
         pid = self._pid(tensor_name)
@@ -116,6 +134,11 @@ class Timeline:
         return True
 
     def end_activity(self, tensor_name: str) -> bool:
+        if self._nvtx is not None:
+            try:
+                self._nvtx.range_pop()
+            except Exception:
+                pass
         if not self._enabled:
             return False
         if tensor_name not in self._active:
