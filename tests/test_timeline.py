@@ -69,6 +69,14 @@ def w_timeline_gpu_spans():
         opt.zero_grad()
         (model(x) ** 2).mean().backward()
         opt.step()
+    # at world size 1 the wrapper skips comm hooks entirely — post comm ops
+    # directly so the GPU lane has spans to show (the span path in
+    # engine.submit is identical at any world size)
+    for i in range(3):
+        h = bf.neighbor_allreduce_nonblocking(
+            torch.ones(1 << 20, device="cuda"), name=f"span{i}"
+        )
+        bf.synchronize(h)
     torch.cuda.synchronize()
     import time
 
