@@ -1246,6 +1246,98 @@ __global__ __launch_bounds__(256) void ln_bwd_finalize_k(
   atomicAdd(&dbeta[col], sb);
 }
 
+// wave-per-row backward (vectorized path): each 64-lane wavefront owns a
+// whole row, so the row loop has NO block barriers — the two row sums
+// reduce with 6 shuffle steps inside the wave. dgamma/dbeta partials fold
+// across the 4 waves through LDS once at kernel end (4-way LDS-atomic
+// conflicts max), then one plain-store scratch row per workgroup.
+template <typename T, int NIT, int VEC>
+__global__ __launch_bounds__(kThreads) void ln_add_bwd_wave_k(
+    T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ r,
+    const T* __restrict__ dy, const float* __restrict__ gamma,
+    const float* __restrict__ mean_s, const float* __restrict__ rstd_s,
+    float* __restrict__ scratch /* [gridDim.x][2H] */, int H, long nrows) {
+  constexpr int kWaves = kThreads / 64;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  extern __shared__ float lds[];  // [2H]
+  float dg[NIT * VEC], db[NIT * VEC];
+#pragma unroll
+  for (int q = 0; q < NIT * VEC; ++q) dg[q] = db[q] = 0.f;
+  for (long row = static_cast<long>(blockIdx.x) * kWaves + wave; row < nrows;
+       row += static_cast<long>(gridDim.x) * kWaves) {
+    const long base = row * H;
+    const float mean = mean_s[row], rstd = rstd_s[row];
+    float xh[NIT * VEC], g[NIT * VEC];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int col = (it * 64 + lane) * VEC;
+      if (col < H) {
+        Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base + col);
+        Pack<T, VEC> rv = *reinterpret_cast<const Pack<T, VEC>*>(r + base + col);
+        Pack<T, VEC> yv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base + col);
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          const int q = it * VEC + u;
+          const float val = AccOf<T>::to(xv.v[u]) + AccOf<T>::to(rv.v[u]);
+          const float xhat = (val - mean) * rstd;
+          const float dyv = AccOf<T>::to(yv.v[u]);
+          const float gg = dyv * gamma[col + u];
+          xh[q] = xhat;
+          g[q] = gg;
+          dg[q] += dyv * xhat;
+          db[q] += dyv;
+          s1 += gg;
+          s2 += gg * xhat;
+        }
+      } else {
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          xh[it * VEC + u] = 0.f;
+          g[it * VEC + u] = 0.f;
+        }
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s1 += __shfl_xor(s1, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
+    const float a1 = s1 / H, a2 = s2 / H;
+#pragma unroll
+    for (int it = 0; it < NIT; ++it) {
+      const int col = (it * 64 + lane) * VEC;
+      if (col < H) {
+        Pack<T, VEC> ov;
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          const int q = it * VEC + u;
+          ov.v[u] = AccOf<T>::from(rstd * (g[q] - a1 - xh[q] * a2));
+        }
+        *reinterpret_cast<Pack<T, VEC>*>(dx + base + col) = ov;
+      }
+    }
+  }
+  for (int c = threadIdx.x; c < 2 * H; c += kThreads) lds[c] = 0.f;
+  __syncthreads();
+#pragma unroll
+  for (int it = 0; it < NIT; ++it) {
+    const int col = (it * 64 + lane) * VEC;
+    if (col < H) {
+#pragma unroll
+      for (int u = 0; u < VEC; ++u) {
+        const int q = it * VEC + u;
+        atomicAdd(&lds[col + u], dg[q]);
+        atomicAdd(&lds[H + col + u], db[q]);
+      }
+    }
+  }
+  __syncthreads();
+  for (int c = threadIdx.x; c < 2 * H; c += kThreads)
+    scratch[static_cast<long>(blockIdx.x) * 2 * H + c] = lds[c];
+}
+
 // pick (NIT, VEC): vectorized 16B packs when every access stays in-row
 // (H % VEC == 0; torch allocations are 256B-aligned so row bases stay
 // 16B-aligned), scalar otherwise. NIT instantiated on {1,2,4} vec /
@@ -1313,14 +1405,44 @@ struct LnAddBwdLauncher {
                         const float* rstd, float* dgamma, float* dbeta,
                         float* scratch, int H, long nrows,
                         hipStream_t stream) {
-    const int grid = ln_bwd_grid(nrows);
+    constexpr int V = vec_width<T>();
+    int grid;
+    // wave path caps at NIT<=4 (4 register arrays of NIT*V floats each;
+    // beyond that the block kernel's lower register footprint wins)
+    if (H % V == 0 && H <= 64 * V * 4 && H <= kThreads * kLnMaxIt) {
+      // wave-per-row kernel: 4 rows in flight per workgroup
+      constexpr int kWaves = kThreads / 64;
+      const long wg = (nrows + kWaves - 1) / kWaves;
+      grid = static_cast<int>(wg < kLnBwdMaxGrid ? (wg > 0 ? wg : 1)
+                                                 : kLnBwdMaxGrid);
+      const size_t lds = 2 * static_cast<size_t>(H) * sizeof(float);
+      const int nit = (H + 64 * V - 1) / (64 * V);
+#define BF_LAUNCH_LN_BWDW(NIT)                                                \
+  hipLaunchKernelGGL((ln_add_bwd_wave_k<T, NIT, V>), dim3(grid),              \
+                     dim3(kThreads), lds, stream, static_cast<T*>(dx),        \
+                     static_cast<const T*>(x), static_cast<const T*>(r),      \
+                     static_cast<const T*>(dy), gamma, mean, rstd, scratch,   \
+                     H, nrows)
+      if (nit <= 1) {
+        BF_LAUNCH_LN_BWDW(1);
+      } else if (nit <= 2) {
+        BF_LAUNCH_LN_BWDW(2);
+      } else if (nit <= 3) {
+        BF_LAUNCH_LN_BWDW(3);
+      } else {
+        BF_LAUNCH_LN_BWDW(4);
+      }
+#undef BF_LAUNCH_LN_BWDW
+    } else {
+      grid = ln_bwd_grid(nrows);
 #define BF_LAUNCH_LN_BWD(NIT, VV)                                             \
   hipLaunchKernelGGL((ln_add_bwd_k<T, NIT, VV>), dim3(grid), dim3(kThreads),  \
                      0, stream, static_cast<T*>(dx), static_cast<const T*>(x),\
                      static_cast<const T*>(r), static_cast<const T*>(dy),     \
                      gamma, mean, rstd, scratch, H, nrows)
-    BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
+      BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
 #undef BF_LAUNCH_LN_BWD
+    }
     BF_CHECK_HIP(hipGetLastError());
     hipLaunchKernelGGL(ln_bwd_finalize_k,
                        dim3(((H + 255) / 256) * kLnFinalizeSlots), dim3(256),
