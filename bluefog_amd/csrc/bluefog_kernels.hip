@@ -416,8 +416,8 @@ __device__ __forceinline__ int bn_channel_of(long i, int C, long HW) {
 // thread at the end (vs 2 per element through LDS).
 template <typename T, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_stats_nhwc_reg_k(
-    const T* __restrict__ x, float* __restrict__ ws /* [2C] */, int C,
-    long numel) {
+    const T* __restrict__ x, float* __restrict__ scratch /* [grid][2C] */,
+    int C, long numel) {
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;  // *VEC elems
   const long nvec = numel / VEC;
   float sum[VEC], sq[VEC];
@@ -447,10 +447,11 @@ __global__ __launch_bounds__(kThreads) void bn_stats_nhwc_reg_k(
     if (sq[v] != 0.f) atomicAdd(&lds[C + c], sq[v]);
   }
   __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (lds[c] != 0.f) atomicAdd(&ws[c], lds[c]);
-    if (lds[C + c] != 0.f) atomicAdd(&ws[C + c], lds[C + c]);
-  }
+  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
+  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
+  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
+    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
 }
 
 template <typename T, int VEC>
@@ -458,7 +459,7 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_nhwc_reg_k(
     const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ ws /* [2C] */, int C, long numel) {
+    float* __restrict__ scratch /* [grid][2C] */, int C, long numel) {
   const long stride = static_cast<long>(gridDim.x) * blockDim.x;
   const long nvec = numel / VEC;
   float sg[VEC], sgx[VEC];
@@ -498,15 +499,16 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_nhwc_reg_k(
     if (sgx[v] != 0.f) atomicAdd(&lds[C + c], sgx[v]);
   }
   __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (lds[c] != 0.f) atomicAdd(&ws[c], lds[c]);
-    if (lds[C + c] != 0.f) atomicAdd(&ws[C + c], lds[C + c]);
-  }
+  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
+  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
+  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
+    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
 }
 
 template <typename T, bool NHWC, int VEC>
 __global__ __launch_bounds__(kThreads) void bn_stats_k(
-    const T* __restrict__ x, float* __restrict__ ws /* [2C]: sum, sumsq */,
+    const T* __restrict__ x, float* __restrict__ scratch /* [grid][2C] */,
     int C, long HW, long numel) {
   extern __shared__ float lds[];  // [2C]
   float* lsum = lds;
@@ -535,10 +537,32 @@ __global__ __launch_bounds__(kThreads) void bn_stats_k(
     atomicAdd(&lsq[c], val * val);
   }
   __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (lsum[c] != 0.f) atomicAdd(&ws[c], lsum[c]);
-    if (lsq[c] != 0.f) atomicAdd(&ws[C + c], lsq[c]);
-  }
+  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
+  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
+  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
+    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
+}
+
+// fold [G][n] per-block partials into ws[n] (+=). Slot dimension split
+// across blocks so small-C reductions are not a 1-block latency pass;
+// kBnReduceSlots atomics per address total.
+constexpr int kBnReduceSlots = 16;
+
+__global__ __launch_bounds__(256) void bn_reduce_slots_k(
+    const float* __restrict__ scratch, float* __restrict__ ws, int n,
+    int G) {
+  const int nchunks = (n + 255) / 256;
+  const int chunk = blockIdx.x % nchunks;
+  const int slice = blockIdx.x / nchunks;
+  const int c = chunk * 256 + threadIdx.x;
+  if (c >= n) return;
+  const int per = (G + kBnReduceSlots - 1) / kBnReduceSlots;
+  const int b0 = slice * per;
+  const int b1 = G < b0 + per ? G : b0 + per;
+  float acc = 0.f;
+  for (int b = b0; b < b1; ++b) acc += scratch[static_cast<long>(b) * n + c];
+  atomicAdd(&ws[c], acc);
 }
 
 __global__ __launch_bounds__(256) void bn_fwd_finalize_k(
@@ -596,7 +620,7 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
     const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ ws /* [2C]: sum_g, sum_g_xhat */, int C, long HW,
+    float* __restrict__ scratch /* [grid][2C] */, int C, long HW,
     long numel) {
   extern __shared__ float lds[];
   float* lg = lds;
@@ -632,10 +656,11 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
     }
   }
   __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    if (lg[c] != 0.f) atomicAdd(&ws[c], lg[c]);
-    if (lgx[c] != 0.f) atomicAdd(&ws[C + c], lgx[c]);
-  }
+  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
+  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
+  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
+  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
+    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
 }
 
 __global__ __launch_bounds__(256) void bn_bwd_finalize_k(
@@ -964,19 +989,29 @@ struct BnFwdLauncher {
                                                static_cast<long>(sgrid) * kThreads * V));
     }
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    float* scratch = nullptr;
+    BF_CHECK_HIP(hipMallocAsync(
+        reinterpret_cast<void**>(&scratch),
+        static_cast<size_t>(sgrid) * 2 * C * sizeof(float), stream));
     if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
       hipLaunchKernelGGL((bn_stats_nhwc_reg_k<T, V>), dim3(sgrid),
                          dim3(kThreads), lds, stream, static_cast<const T*>(x),
-                         ws, C, numel);
+                         scratch, C, numel);
     } else {
 #define BF_BN_STATS(NH, VV)                                                    \
   hipLaunchKernelGGL((bn_stats_k<T, NH, VV>), dim3(sgrid), dim3(kThreads),     \
-                     lds, stream, static_cast<const T*>(x), ws, C, HW, numel)
+                     lds, stream, static_cast<const T*>(x), scratch, C, HW,    \
+                     numel)
       if (nhwc) { if (vec) BF_BN_STATS(true, V); else BF_BN_STATS(true, 1); }
       else      { if (vec) BF_BN_STATS(false, V); else BF_BN_STATS(false, 1); }
 #undef BF_BN_STATS
     }
     BF_CHECK_HIP(hipGetLastError());
+    hipLaunchKernelGGL(bn_reduce_slots_k,
+                       dim3(((2 * C + 255) / 256) * kBnReduceSlots), dim3(256),
+                       0, stream, scratch, ws, 2 * C, sgrid);
+    BF_CHECK_HIP(hipGetLastError());
+    BF_CHECK_HIP(hipFreeAsync(scratch, stream));
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_fwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
                        stream, ws, run_mean_in, run_var_in, run_mean, run_var,
@@ -1013,22 +1048,31 @@ struct BnBwdLauncher {
       sgrid *= C / static_cast<int>(std::gcd(static_cast<long>(C),
                                              static_cast<long>(sgrid) * kThreads * V));
     const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
+    float* scratch = nullptr;
+    BF_CHECK_HIP(hipMallocAsync(
+        reinterpret_cast<void**>(&scratch),
+        static_cast<size_t>(sgrid) * 2 * C * sizeof(float), stream));
     if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
       hipLaunchKernelGGL((bn_bwd_reduce_nhwc_reg_k<T, V>), dim3(sgrid),
                          dim3(kThreads), lds, stream, static_cast<const T*>(x),
                          static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, ws, C, numel);
+                         save_invstd, scratch, C, numel);
     } else {
 #define BF_BN_RED(NH, VV)                                                      \
   hipLaunchKernelGGL((bn_bwd_reduce_k<T, NH, VV>), dim3(sgrid),                \
                      dim3(kThreads), lds, stream, static_cast<const T*>(x),    \
                      static_cast<const T*>(dy), gamma, beta, save_mean,        \
-                     save_invstd, ws, C, HW, numel)
+                     save_invstd, scratch, C, HW, numel)
       if (nhwc) { if (vec) BF_BN_RED(true, V); else BF_BN_RED(true, 1); }
       else      { if (vec) BF_BN_RED(false, V); else BF_BN_RED(false, 1); }
 #undef BF_BN_RED
     }
     BF_CHECK_HIP(hipGetLastError());
+    hipLaunchKernelGGL(bn_reduce_slots_k,
+                       dim3(((2 * C + 255) / 256) * kBnReduceSlots), dim3(256),
+                       0, stream, scratch, ws, 2 * C, sgrid);
+    BF_CHECK_HIP(hipGetLastError());
+    BF_CHECK_HIP(hipFreeAsync(scratch, stream));
     const float count = static_cast<float>(numel / C);
     hipLaunchKernelGGL(bn_bwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
                        stream, ws, gamma, save_invstd, dgamma, dbeta, coef,
