@@ -146,39 +146,6 @@ def run(numel, iters, dtype, dev):
         )
         add("ln_add_bwd[16x512x768]", ms, 4 * nl * esz)
 
-    if dtype == torch.float32:
-        # fused BN+ReLU pipeline (opt-in path) on a ResNet50-stem-shaped
-        # activation: measures how far the reductions are from peak
-        N, C, H, W = 64, 64, 112, 112
-        xbn = torch.randn(N, C, H, W, device=dev).to(memory_format=torch.channels_last)
-        ybn = torch.empty_like(xbn)
-        dxbn = torch.empty_like(xbn)
-        gamma = torch.rand(C, device=dev) + 0.5
-        beta = torch.randn(C, device=dev)
-        rm = torch.zeros(C, device=dev)
-        rv = torch.ones(C, device=dev)
-        sm = torch.empty(C, device=dev)
-        si = torch.empty(C, device=dev)
-        ws_bn = torch.zeros(2 * C, device=dev)
-        coef = torch.empty(3 * C, device=dev)
-        dgamma = torch.empty(C, device=dev)
-        dbeta = torch.empty(C, device=dev)
-        nbn = xbn.numel()
-        ms = time_kernel(
-            lambda: _C.bn_relu_fwd(ybn, xbn, gamma, beta, rm, rv, sm, si, ws_bn,
-                                   0.1, 1e-5),
-            iters,
-        )
-        rows.append({"kernel": "bn_relu_fwd[64x64x112x112 nhwc]", "dtype": "float32",
-                     "numel": nbn, "ms": ms, "GBps": 2 * nbn * 4 / ms / 1e6})
-        ms = time_kernel(
-            lambda: _C.bn_relu_bwd(dxbn, xbn, ybn, gamma, beta, sm, si, ws_bn,
-                                   dgamma, dbeta, coef),
-            iters,
-        )
-        rows.append({"kernel": "bn_relu_bwd[64x64x112x112 nhwc]", "dtype": "float32",
-                     "numel": nbn, "ms": ms, "GBps": 5 * nbn * 4 / ms / 1e6})
-
     ms = time_kernel(lambda: hip_ext.scale_put(out, self_t, 0.25), iters)
     add("scale_put", ms, 2 * numel * esz)
     ms = time_kernel(lambda: hip_ext.accum_put(out, self_t, 0.25), iters)

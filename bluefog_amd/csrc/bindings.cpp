@@ -37,18 +37,6 @@ hipError_t bf_add_relu_fwd(void* out, const void* a, const void* b, long numel,
                            int dtype, hipStream_t stream);
 hipError_t bf_relu_bwd_mask(void* gin, const void* g, const void* out,
                             long numel, int dtype, hipStream_t stream);
-hipError_t bf_bn_relu_fwd(void* y, const void* x, const float* gamma,
-                          const float* beta, const float* run_mean_in,
-                          const float* run_var_in, float* run_mean,
-                          float* run_var, float* save_mean, float* save_invstd,
-                          float* ws, float momentum, float eps, int C, long HW,
-                          long numel, bool nhwc, int dtype, hipStream_t stream);
-hipError_t bf_bn_relu_bwd(void* dx, const void* x, const void* dy,
-                          const float* gamma, const float* beta,
-                          const float* save_mean, const float* save_invstd,
-                          float* ws, float* dgamma, float* dbeta, float* coef,
-                          int C, long HW, long numel, bool nhwc, int dtype,
-                          hipStream_t stream);
 hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,
                          const float* gamma, const float* beta, float* mean,
                          float* rstd, int H, long nrows, double eps, int dtype,
@@ -206,73 +194,6 @@ void relu_bwd_mask(at::Tensor gin, at::Tensor g, at::Tensor out) {
             "relu_bwd_mask");
 }
 
-bool bn_layout(const at::Tensor& t, bool& nhwc) {
-  // dense NCHW-contiguous or channels_last 4-D
-  if (t.dim() != 4) return false;
-  if (t.is_contiguous(at::MemoryFormat::ChannelsLast)) {
-    nhwc = true;
-    return true;
-  }
-  if (t.is_contiguous()) {
-    nhwc = false;
-    return true;
-  }
-  return false;
-}
-
-void bn_relu_fwd(at::Tensor y, at::Tensor x, at::Tensor gamma, at::Tensor beta,
-                 at::Tensor running_mean, at::Tensor running_var,
-                 at::Tensor save_mean, at::Tensor save_invstd, at::Tensor ws,
-                 double momentum, double eps) {
-  bool nhwc = false;
-  TORCH_CHECK(bn_layout(x, nhwc), "bn_relu_fwd: x must be dense NCHW or channels_last");
-  bool nhwc_y = false;
-  TORCH_CHECK(bn_layout(y, nhwc_y) && nhwc_y == nhwc && y.sizes() == x.sizes(),
-              "bn_relu_fwd: y layout must match x");
-  TORCH_CHECK(x.scalar_type() != at::kDouble, "bn_relu_fwd: f64 unsupported");
-  const int C = static_cast<int>(x.size(1));
-  const long HW = x.size(2) * x.size(3);
-  for (const auto& t : {gamma, beta, running_mean, running_var, save_mean, save_invstd})
-    TORCH_CHECK(t.scalar_type() == at::kFloat && t.numel() == C && t.is_contiguous(),
-                "bn_relu_fwd: per-channel tensors must be fp32[C]");
-  TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.numel() >= 2 * C, "ws must be fp32[2C]");
-  check_hip(bf_bn_relu_fwd(y.data_ptr(), x.data_ptr(), gamma.data_ptr<float>(),
-                           beta.data_ptr<float>(), running_mean.data_ptr<float>(),
-                           running_var.data_ptr<float>(),
-                           running_mean.data_ptr<float>(),
-                           running_var.data_ptr<float>(),
-                           save_mean.data_ptr<float>(),
-                           save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
-                           static_cast<float>(momentum), static_cast<float>(eps),
-                           C, HW, x.numel(), nhwc, dtype_code(x),
-                           current_stream()),
-            "bn_relu_fwd");
-}
-
-void bn_relu_bwd(at::Tensor dx, at::Tensor x, at::Tensor dy, at::Tensor gamma,
-                 at::Tensor beta, at::Tensor save_mean, at::Tensor save_invstd,
-                 at::Tensor ws, at::Tensor dgamma, at::Tensor dbeta,
-                 at::Tensor coef) {
-  bool nhwc = false;
-  TORCH_CHECK(bn_layout(x, nhwc), "bn_relu_bwd: x must be dense NCHW or channels_last");
-  bool l2 = false, l3 = false;
-  TORCH_CHECK(bn_layout(dy, l2) && l2 == nhwc && bn_layout(dx, l3) && l3 == nhwc,
-              "bn_relu_bwd: dy/dx layout must match x");
-  const int C = static_cast<int>(x.size(1));
-  const long HW = x.size(2) * x.size(3);
-  TORCH_CHECK(coef.scalar_type() == at::kFloat && coef.numel() >= 3 * C, "coef fp32[3C]");
-  TORCH_CHECK(ws.scalar_type() == at::kFloat && ws.numel() >= 2 * C, "ws fp32[2C]");
-  check_hip(bf_bn_relu_bwd(dx.data_ptr(), x.data_ptr(), dy.data_ptr(),
-                           gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                           save_mean.data_ptr<float>(),
-                           save_invstd.data_ptr<float>(), ws.data_ptr<float>(),
-                           dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                           coef.data_ptr<float>(), C, HW, x.numel(), nhwc,
-                           dtype_code(x), current_stream()),
-            "bn_relu_bwd");
-}
-
-
 void ln_add_fwd(at::Tensor y, at::Tensor x, at::Tensor r, at::Tensor gamma,
                 at::Tensor beta, at::Tensor mean, at::Tensor rstd, double eps) {
   TORCH_CHECK(x.is_contiguous() && r.is_contiguous() && y.is_contiguous(),
@@ -336,10 +257,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused neighbor-average + Adam over a flat bucket");
   m.def("add_relu_fwd", &add_relu_fwd, "out = max(a+b, 0)");
   m.def("relu_bwd_mask", &relu_bwd_mask, "gin = out>0 ? g : 0");
-  m.def("bn_relu_fwd", &bn_relu_fwd,
-        "fused training BatchNorm2d+ReLU forward (stats + normalize)");
-  m.def("bn_relu_bwd", &bn_relu_bwd,
-        "fused BatchNorm2d+ReLU backward (reduce + dx, mask from x)");
   m.def("ln_add_fwd", &ln_add_fwd,
         "y = LayerNorm(x + r); saves per-row mean/rstd");
   m.def("ln_add_bwd", &ln_add_bwd,

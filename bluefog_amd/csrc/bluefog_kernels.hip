@@ -390,330 +390,6 @@ __global__ __launch_bounds__(kThreads) void relu_bwd_mask_k(
     gin[t] = AccOf<T>::to(out[t]) > static_cast<Acc>(0) ? g[t] : AccOf<T>::from(0.f);
 }
 
-// ---------------------------------------------------------------------------
-// Fused training BatchNorm2d + ReLU (NHWC or NCHW dense layouts).
-//
-// Replaces the MIOpen BN kernel chain + separate ReLU clamp/threshold
-// kernels in the ResNet hot path with minimal-traffic passes:
-//   fwd:  stats (1 read of x)  ->  finalize (C)  ->  normalize+ReLU (1R 1W)
-//   bwd:  reduce (x,dy: 2R)    ->  finalize (C)  ->  dx (x,dy: 2R 1W)
-// The ReLU mask is recomputed from x (gamma*xhat+beta > 0), so neither the
-// forward output nor a mask tensor is re-read in backward.
-// Per-channel accumulation: one LDS array per workgroup, one global
-// atomicAdd per (channel, workgroup) — C <= kMaxBnChannels floats of LDS.
-// ---------------------------------------------------------------------------
-
-constexpr int kMaxBnChannels = 4096;  // 2 fp32 arrays * 4096 = 32 KB LDS
-
-template <typename T, bool NHWC>
-__device__ __forceinline__ int bn_channel_of(long i, int C, long HW) {
-  return NHWC ? static_cast<int>(i % C) : static_cast<int>((i / HW) % C);
-}
-
-// NHWC fast path: the launcher sizes the grid so that the grid stride is a
-// multiple of C — each lane then touches a FIXED set of VEC channels for
-// its whole loop and accumulates in registers; only 2*VEC atomics per
-// thread at the end (vs 2 per element through LDS).
-template <typename T, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_stats_nhwc_reg_k(
-    const T* __restrict__ x, float* __restrict__ scratch /* [grid][2C] */,
-    int C, long numel) {
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;  // *VEC elems
-  const long nvec = numel / VEC;
-  float sum[VEC], sq[VEC];
-#pragma unroll
-  for (int v = 0; v < VEC; ++v) sum[v] = sq[v] = 0.f;
-  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int c0 = static_cast<int>((i * VEC) % C);  // fixed: stride*VEC % C == 0
-  for (; i < nvec; i += stride) {
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + i * VEC);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const float val = AccOf<T>::to(xv.v[v]);
-      sum[v] += val;
-      sq[v] += val * val;
-    }
-  }
-  // workgroup LDS reduction of the per-lane totals, then one global
-  // atomic per channel per workgroup (global contention ~ gridDim, not
-  // ~ elements/threads)
-  extern __shared__ float lds[];  // [2C]
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
-  __syncthreads();
-#pragma unroll
-  for (int v = 0; v < VEC; ++v) {
-    const int c = (c0 + v) % C;
-    if (sum[v] != 0.f) atomicAdd(&lds[c], sum[v]);
-    if (sq[v] != 0.f) atomicAdd(&lds[C + c], sq[v]);
-  }
-  __syncthreads();
-  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
-  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
-  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
-    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
-}
-
-template <typename T, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_bwd_reduce_nhwc_reg_k(
-    const T* __restrict__ x, const T* __restrict__ dy,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ scratch /* [grid][2C] */, int C, long numel) {
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  const long nvec = numel / VEC;
-  float sg[VEC], sgx[VEC];
-#pragma unroll
-  for (int v = 0; v < VEC; ++v) sg[v] = sgx[v] = 0.f;
-  long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-  const int c0 = static_cast<int>((i * VEC) % C);
-  float mv[VEC], iv[VEC], gm[VEC], bt[VEC];
-#pragma unroll
-  for (int v = 0; v < VEC; ++v) {
-    const int c = (c0 + v) % C;
-    mv[v] = mean[c];
-    iv[v] = invstd[c];
-    gm[v] = gamma[c];
-    bt[v] = beta[c];
-  }
-  for (; i < nvec; i += stride) {
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + i * VEC);
-    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + i * VEC);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const float xhat = (AccOf<T>::to(xv.v[v]) - mv[v]) * iv[v];
-      if (gm[v] * xhat + bt[v] > 0.f) {
-        const float g = AccOf<T>::to(gv.v[v]);
-        sg[v] += g;
-        sgx[v] += g * xhat;
-      }
-    }
-  }
-  extern __shared__ float lds[];  // [2C]
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
-  __syncthreads();
-#pragma unroll
-  for (int v = 0; v < VEC; ++v) {
-    const int c = (c0 + v) % C;
-    if (sg[v] != 0.f) atomicAdd(&lds[c], sg[v]);
-    if (sgx[v] != 0.f) atomicAdd(&lds[C + c], sgx[v]);
-  }
-  __syncthreads();
-  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
-  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
-  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
-    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
-}
-
-template <typename T, bool NHWC, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_stats_k(
-    const T* __restrict__ x, float* __restrict__ scratch /* [grid][2C] */,
-    int C, long HW, long numel) {
-  extern __shared__ float lds[];  // [2C]
-  float* lsum = lds;
-  float* lsq = lds + C;
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
-  __syncthreads();
-  const long nvec = numel / VEC;
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < nvec; i += stride) {
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const float val = AccOf<T>::to(xv.v[v]);
-      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
-      atomicAdd(&lsum[c], val);
-      atomicAdd(&lsq[c], val * val);
-    }
-  }
-  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
-       t < numel; t += stride) {
-    const float val = AccOf<T>::to(x[t]);
-    const int c = bn_channel_of<T, NHWC>(t, C, HW);
-    atomicAdd(&lsum[c], val);
-    atomicAdd(&lsq[c], val * val);
-  }
-  __syncthreads();
-  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
-  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
-  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
-    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
-}
-
-// fold [G][n] per-block partials into ws[n] (+=). Slot dimension split
-// across blocks so small-C reductions are not a 1-block latency pass;
-// kBnReduceSlots atomics per address total.
-constexpr int kBnReduceSlots = 16;
-
-__global__ __launch_bounds__(256) void bn_reduce_slots_k(
-    const float* __restrict__ scratch, float* __restrict__ ws, int n,
-    int G) {
-  const int nchunks = (n + 255) / 256;
-  const int chunk = blockIdx.x % nchunks;
-  const int slice = blockIdx.x / nchunks;
-  const int c = chunk * 256 + threadIdx.x;
-  if (c >= n) return;
-  const int per = (G + kBnReduceSlots - 1) / kBnReduceSlots;
-  const int b0 = slice * per;
-  const int b1 = G < b0 + per ? G : b0 + per;
-  float acc = 0.f;
-  for (int b = b0; b < b1; ++b) acc += scratch[static_cast<long>(b) * n + c];
-  atomicAdd(&ws[c], acc);
-}
-
-__global__ __launch_bounds__(256) void bn_fwd_finalize_k(
-    float* __restrict__ ws, const float* __restrict__ running_mean_in,
-    const float* __restrict__ running_var_in, float* __restrict__ running_mean,
-    float* __restrict__ running_var, float* __restrict__ save_mean,
-    float* __restrict__ save_invstd, float momentum, float eps, float count,
-    int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float mean = ws[c] / count;
-  float var = ws[C + c] / count - mean * mean;
-  if (var < 0.f) var = 0.f;
-  ws[c] = 0.f;      // reset for the next launch (module reuses the buffer)
-  ws[C + c] = 0.f;
-  save_mean[c] = mean;
-  save_invstd[c] = rsqrtf(var + eps);
-  const float unbiased = count > 1.f ? var * count / (count - 1.f) : var;
-  running_mean[c] = (1.f - momentum) * running_mean_in[c] + momentum * mean;
-  running_var[c] = (1.f - momentum) * running_var_in[c] + momentum * unbiased;
-}
-
-template <typename T, bool NHWC, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_fwd_norm_relu_k(
-    T* __restrict__ y, const T* __restrict__ x,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ mean, const float* __restrict__ invstd, int C,
-    long HW, long numel) {
-  const long nvec = numel / VEC;
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < nvec; i += stride) {
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
-      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
-      const float val = gamma[c] * xhat + beta[c];
-      xv.v[v] = AccOf<T>::from(val > 0.f ? val : 0.f);
-    }
-    *reinterpret_cast<Pack<T, VEC>*>(y + base) = xv;
-  }
-  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
-       t < numel; t += stride) {
-    const int c = bn_channel_of<T, NHWC>(t, C, HW);
-    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
-    float v = gamma[c] * xhat + beta[c];
-    y[t] = AccOf<T>::from(v > 0.f ? v : 0.f);
-  }
-}
-
-template <typename T, bool NHWC, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_bwd_reduce_k(
-    const T* __restrict__ x, const T* __restrict__ dy,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ scratch /* [grid][2C] */, int C, long HW,
-    long numel) {
-  extern __shared__ float lds[];
-  float* lg = lds;
-  float* lgx = lds + C;
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x) lds[c] = 0.f;
-  __syncthreads();
-  const long nvec = numel / VEC;
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < nvec; i += stride) {
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
-    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
-      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
-      if (gamma[c] * xhat + beta[c] > 0.f) {
-        const float g = AccOf<T>::to(gv.v[v]);
-        atomicAdd(&lg[c], g);
-        atomicAdd(&lgx[c], g * xhat);
-      }
-    }
-  }
-  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
-       t < numel; t += stride) {
-    const int c = bn_channel_of<T, NHWC>(t, C, HW);
-    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
-    if (gamma[c] * xhat + beta[c] > 0.f) {
-      const float g = AccOf<T>::to(dy[t]);
-      atomicAdd(&lg[c], g);
-      atomicAdd(&lgx[c], g * xhat);
-    }
-  }
-  __syncthreads();
-  // per-block partials as PLAIN stores; reduce_slots_k folds [grid][2C]
-  // into ws afterwards (grid-fold atomics to 2C addresses were the 2x
-  // BN-vs-MIOpen end-to-end loss — same disease as the LN backward)
-  for (int c = threadIdx.x; c < 2 * C; c += blockDim.x)
-    scratch[static_cast<long>(blockIdx.x) * 2 * C + c] = lds[c];
-}
-
-__global__ __launch_bounds__(256) void bn_bwd_finalize_k(
-    float* __restrict__ ws, const float* __restrict__ gamma,
-    const float* __restrict__ invstd, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, float* __restrict__ coef /* [3C]: a,b,c */,
-    float count, int C) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const float sum_g = ws[c];
-  const float sum_gx = ws[C + c];
-  ws[c] = 0.f;
-  ws[C + c] = 0.f;
-  dbeta[c] = sum_g;
-  dgamma[c] = sum_gx;
-  coef[c] = gamma[c] * invstd[c];           // a
-  coef[C + c] = sum_g / count;              // b (mean of g)
-  coef[2 * C + c] = sum_gx / count;         // c (mean of g*xhat)
-}
-
-template <typename T, bool NHWC, int VEC>
-__global__ __launch_bounds__(kThreads) void bn_bwd_dx_k(
-    T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ dy,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ coef, int C, long HW, long numel) {
-  const long nvec = numel / VEC;
-  const long stride = static_cast<long>(gridDim.x) * blockDim.x;
-  for (long i = static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x;
-       i < nvec; i += stride) {
-    const long base = i * VEC;
-    Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base);
-    Pack<T, VEC> gv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base);
-#pragma unroll
-    for (int v = 0; v < VEC; ++v) {
-      const int c = bn_channel_of<T, NHWC>(base + v, C, HW);
-      const float xhat = (AccOf<T>::to(xv.v[v]) - mean[c]) * invstd[c];
-      const float g =
-          gamma[c] * xhat + beta[c] > 0.f ? AccOf<T>::to(gv.v[v]) : 0.f;
-      xv.v[v] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
-    }
-    *reinterpret_cast<Pack<T, VEC>*>(dx + base) = xv;
-  }
-  for (long t = nvec * VEC + (static_cast<long>(blockIdx.x) * blockDim.x + threadIdx.x);
-       t < numel; t += stride) {
-    const int c = bn_channel_of<T, NHWC>(t, C, HW);
-    const float xhat = (AccOf<T>::to(x[t]) - mean[c]) * invstd[c];
-    const float pre = gamma[c] * xhat + beta[c];
-    const float g = pre > 0.f ? AccOf<T>::to(dy[t]) : 0.f;
-    dx[t] = AccOf<T>::from(coef[c] * (g - coef[C + c] - xhat * coef[2 * C + c]));
-  }
-}
-
 enum BfDtype : int { kF32 = 0, kF64 = 1, kF16 = 2, kBF16 = 3 };
 
 template <template <typename> class Fn, typename... Args>
@@ -959,139 +635,6 @@ struct ReluBwdMaskLauncher {
   }
 };
 
-inline int bn_grid(long numel) {
-  long blocks = (numel + kThreads * 64 - 1) / (kThreads * 64);
-  if (blocks > 1024) blocks = 1024;
-  if (blocks < 8) blocks = 8;
-  return static_cast<int>(blocks);
-}
-
-template <typename T>
-struct BnFwdLauncher {
-  static hipError_t run(void* y, const void* x, const float* gamma,
-                        const float* beta, const float* run_mean_in,
-                        const float* run_var_in, float* run_mean,
-                        float* run_var, float* save_mean, float* save_invstd,
-                        float* ws, float momentum, float eps, int C, long HW,
-                        long numel, bool nhwc, hipStream_t stream) {
-    if (C > kMaxBnChannels) return hipErrorInvalidValue;  // LDS budget
-    constexpr int V = vec_width<T>();
-    const bool vec = vec_ok<T>(x, numel);
-    int sgrid = bn_grid(numel / (vec ? V : 1));
-    // register path: fixed channel per lane requires (grid*threads*V) % C == 0
-    const bool reg = nhwc && vec && (C % V == 0);
-    if (reg) {
-      const int chunks = (C + kThreads * V - 1) / (kThreads * V);
-      const int align = chunks > 1 ? chunks : 1;  // C > lane coverage
-      sgrid = ((sgrid + align - 1) / align) * align;
-      if ((static_cast<long>(sgrid) * kThreads * V) % C != 0)
-        sgrid *= C / static_cast<int>(std::gcd(static_cast<long>(C),
-                                               static_cast<long>(sgrid) * kThreads * V));
-    }
-    const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
-    float* scratch = nullptr;
-    BF_CHECK_HIP(hipMallocAsync(
-        reinterpret_cast<void**>(&scratch),
-        static_cast<size_t>(sgrid) * 2 * C * sizeof(float), stream));
-    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
-      hipLaunchKernelGGL((bn_stats_nhwc_reg_k<T, V>), dim3(sgrid),
-                         dim3(kThreads), lds, stream, static_cast<const T*>(x),
-                         scratch, C, numel);
-    } else {
-#define BF_BN_STATS(NH, VV)                                                    \
-  hipLaunchKernelGGL((bn_stats_k<T, NH, VV>), dim3(sgrid), dim3(kThreads),     \
-                     lds, stream, static_cast<const T*>(x), scratch, C, HW,    \
-                     numel)
-      if (nhwc) { if (vec) BF_BN_STATS(true, V); else BF_BN_STATS(true, 1); }
-      else      { if (vec) BF_BN_STATS(false, V); else BF_BN_STATS(false, 1); }
-#undef BF_BN_STATS
-    }
-    BF_CHECK_HIP(hipGetLastError());
-    hipLaunchKernelGGL(bn_reduce_slots_k,
-                       dim3(((2 * C + 255) / 256) * kBnReduceSlots), dim3(256),
-                       0, stream, scratch, ws, 2 * C, sgrid);
-    BF_CHECK_HIP(hipGetLastError());
-    BF_CHECK_HIP(hipFreeAsync(scratch, stream));
-    const float count = static_cast<float>(numel / C);
-    hipLaunchKernelGGL(bn_fwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
-                       stream, ws, run_mean_in, run_var_in, run_mean, run_var,
-                       save_mean, save_invstd, momentum, eps, count, C);
-    BF_CHECK_HIP(hipGetLastError());
-    const bool nvecok = vec && vec_ok<T>(y, numel);
-    const int ngrid = grid_for((numel + (nvecok ? V : 1) - 1) / (nvecok ? V : 1));
-#define BF_BN_NORM(NH, VV)                                                     \
-  hipLaunchKernelGGL((bn_fwd_norm_relu_k<T, NH, VV>), dim3(ngrid),             \
-                     dim3(kThreads), 0, stream, static_cast<T*>(y),            \
-                     static_cast<const T*>(x), gamma, beta, save_mean,         \
-                     save_invstd, C, HW, numel)
-    if (nhwc) { if (nvecok) BF_BN_NORM(true, V); else BF_BN_NORM(true, 1); }
-    else      { if (nvecok) BF_BN_NORM(false, V); else BF_BN_NORM(false, 1); }
-#undef BF_BN_NORM
-    return hipGetLastError();
-  }
-};
-
-template <typename T>
-struct BnBwdLauncher {
-  static hipError_t run(void* dx, const void* x, const void* dy,
-                        const float* gamma, const float* beta,
-                        const float* save_mean, const float* save_invstd,
-                        float* ws, float* dgamma, float* dbeta, float* coef,
-                        int C, long HW, long numel, bool nhwc,
-                        hipStream_t stream) {
-    if (C > kMaxBnChannels) return hipErrorInvalidValue;  // LDS budget
-    constexpr int V = vec_width<T>();
-    const bool vec = vec_ok<T>(x, numel) && vec_ok<T>(dy, numel);
-    int sgrid = bn_grid(numel / (vec ? V : 1));
-    const bool reg = nhwc && vec && (C % V == 0);
-    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C != 0)
-      sgrid *= C / static_cast<int>(std::gcd(static_cast<long>(C),
-                                             static_cast<long>(sgrid) * kThreads * V));
-    const size_t lds = 2 * static_cast<size_t>(C) * sizeof(float);
-    float* scratch = nullptr;
-    BF_CHECK_HIP(hipMallocAsync(
-        reinterpret_cast<void**>(&scratch),
-        static_cast<size_t>(sgrid) * 2 * C * sizeof(float), stream));
-    if (reg && (static_cast<long>(sgrid) * kThreads * V) % C == 0) {
-      hipLaunchKernelGGL((bn_bwd_reduce_nhwc_reg_k<T, V>), dim3(sgrid),
-                         dim3(kThreads), lds, stream, static_cast<const T*>(x),
-                         static_cast<const T*>(dy), gamma, beta, save_mean,
-                         save_invstd, scratch, C, numel);
-    } else {
-#define BF_BN_RED(NH, VV)                                                      \
-  hipLaunchKernelGGL((bn_bwd_reduce_k<T, NH, VV>), dim3(sgrid),                \
-                     dim3(kThreads), lds, stream, static_cast<const T*>(x),    \
-                     static_cast<const T*>(dy), gamma, beta, save_mean,        \
-                     save_invstd, scratch, C, HW, numel)
-      if (nhwc) { if (vec) BF_BN_RED(true, V); else BF_BN_RED(true, 1); }
-      else      { if (vec) BF_BN_RED(false, V); else BF_BN_RED(false, 1); }
-#undef BF_BN_RED
-    }
-    BF_CHECK_HIP(hipGetLastError());
-    hipLaunchKernelGGL(bn_reduce_slots_k,
-                       dim3(((2 * C + 255) / 256) * kBnReduceSlots), dim3(256),
-                       0, stream, scratch, ws, 2 * C, sgrid);
-    BF_CHECK_HIP(hipGetLastError());
-    BF_CHECK_HIP(hipFreeAsync(scratch, stream));
-    const float count = static_cast<float>(numel / C);
-    hipLaunchKernelGGL(bn_bwd_finalize_k, dim3((C + 255) / 256), dim3(256), 0,
-                       stream, ws, gamma, save_invstd, dgamma, dbeta, coef,
-                       count, C);
-    BF_CHECK_HIP(hipGetLastError());
-    const bool dvec = vec && vec_ok<T>(dx, numel);
-    const int ngrid = grid_for((numel + (dvec ? V : 1) - 1) / (dvec ? V : 1));
-#define BF_BN_DX(NH, VV)                                                       \
-  hipLaunchKernelGGL((bn_bwd_dx_k<T, NH, VV>), dim3(ngrid), dim3(kThreads), 0, \
-                     stream, static_cast<T*>(dx), static_cast<const T*>(x),    \
-                     static_cast<const T*>(dy), gamma, beta, save_mean,        \
-                     save_invstd, coef, C, HW, numel)
-    if (nhwc) { if (dvec) BF_BN_DX(true, V); else BF_BN_DX(true, 1); }
-    else      { if (dvec) BF_BN_DX(false, V); else BF_BN_DX(false, 1); }
-#undef BF_BN_DX
-    return hipGetLastError();
-  }
-};
-
 // ---------------------------------------------------------------------------
 // fused residual add + LayerNorm (BERT hot path): y = LN(x + r) over the
 // last dimension H. One workgroup (4 waves) per row, grid-strided over
@@ -1290,98 +833,6 @@ __global__ __launch_bounds__(256) void ln_bwd_finalize_k(
   atomicAdd(&dbeta[col], sb);
 }
 
-// wave-per-row backward (vectorized path): each 64-lane wavefront owns a
-// whole row, so the row loop has NO block barriers — the two row sums
-// reduce with 6 shuffle steps inside the wave. dgamma/dbeta partials fold
-// across the 4 waves through LDS once at kernel end (4-way LDS-atomic
-// conflicts max), then one plain-store scratch row per workgroup.
-template <typename T, int NIT, int VEC>
-__global__ __launch_bounds__(kThreads) void ln_add_bwd_wave_k(
-    T* __restrict__ dx, const T* __restrict__ x, const T* __restrict__ r,
-    const T* __restrict__ dy, const float* __restrict__ gamma,
-    const float* __restrict__ mean_s, const float* __restrict__ rstd_s,
-    float* __restrict__ scratch /* [gridDim.x][2H] */, int H, long nrows) {
-  constexpr int kWaves = kThreads / 64;
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  extern __shared__ float lds[];  // [2H]
-  float dg[NIT * VEC], db[NIT * VEC];
-#pragma unroll
-  for (int q = 0; q < NIT * VEC; ++q) dg[q] = db[q] = 0.f;
-  for (long row = static_cast<long>(blockIdx.x) * kWaves + wave; row < nrows;
-       row += static_cast<long>(gridDim.x) * kWaves) {
-    const long base = row * H;
-    const float mean = mean_s[row], rstd = rstd_s[row];
-    float xh[NIT * VEC], g[NIT * VEC];
-    float s1 = 0.f, s2 = 0.f;
-#pragma unroll
-    for (int it = 0; it < NIT; ++it) {
-      const int col = (it * 64 + lane) * VEC;
-      if (col < H) {
-        Pack<T, VEC> xv = *reinterpret_cast<const Pack<T, VEC>*>(x + base + col);
-        Pack<T, VEC> rv = *reinterpret_cast<const Pack<T, VEC>*>(r + base + col);
-        Pack<T, VEC> yv = *reinterpret_cast<const Pack<T, VEC>*>(dy + base + col);
-#pragma unroll
-        for (int u = 0; u < VEC; ++u) {
-          const int q = it * VEC + u;
-          const float val = AccOf<T>::to(xv.v[u]) + AccOf<T>::to(rv.v[u]);
-          const float xhat = (val - mean) * rstd;
-          const float dyv = AccOf<T>::to(yv.v[u]);
-          const float gg = dyv * gamma[col + u];
-          xh[q] = xhat;
-          g[q] = gg;
-          dg[q] += dyv * xhat;
-          db[q] += dyv;
-          s1 += gg;
-          s2 += gg * xhat;
-        }
-      } else {
-#pragma unroll
-        for (int u = 0; u < VEC; ++u) {
-          xh[it * VEC + u] = 0.f;
-          g[it * VEC + u] = 0.f;
-        }
-      }
-    }
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      s1 += __shfl_xor(s1, off, 64);
-      s2 += __shfl_xor(s2, off, 64);
-    }
-    const float a1 = s1 / H, a2 = s2 / H;
-#pragma unroll
-    for (int it = 0; it < NIT; ++it) {
-      const int col = (it * 64 + lane) * VEC;
-      if (col < H) {
-        Pack<T, VEC> ov;
-#pragma unroll
-        for (int u = 0; u < VEC; ++u) {
-          const int q = it * VEC + u;
-          ov.v[u] = AccOf<T>::from(rstd * (g[q] - a1 - xh[q] * a2));
-        }
-        *reinterpret_cast<Pack<T, VEC>*>(dx + base + col) = ov;
-      }
-    }
-  }
-  for (int c = threadIdx.x; c < 2 * H; c += kThreads) lds[c] = 0.f;
-  __syncthreads();
-#pragma unroll
-  for (int it = 0; it < NIT; ++it) {
-    const int col = (it * 64 + lane) * VEC;
-    if (col < H) {
-#pragma unroll
-      for (int u = 0; u < VEC; ++u) {
-        const int q = it * VEC + u;
-        atomicAdd(&lds[col + u], dg[q]);
-        atomicAdd(&lds[H + col + u], db[q]);
-      }
-    }
-  }
-  __syncthreads();
-  for (int c = threadIdx.x; c < 2 * H; c += kThreads)
-    scratch[static_cast<long>(blockIdx.x) * 2 * H + c] = lds[c];
-}
-
 // pick (NIT, VEC): vectorized 16B packs when every access stays in-row
 // (H % VEC == 0; torch allocations are 256B-aligned so row bases stay
 // 16B-aligned), scalar otherwise. NIT instantiated on {1,2,4} vec /
@@ -1449,44 +900,14 @@ struct LnAddBwdLauncher {
                         const float* rstd, float* dgamma, float* dbeta,
                         float* scratch, int H, long nrows,
                         hipStream_t stream) {
-    constexpr int V = vec_width<T>();
-    int grid;
-    // wave path caps at NIT<=4 (4 register arrays of NIT*V floats each;
-    // beyond that the block kernel's lower register footprint wins)
-    if (H % V == 0 && H <= 64 * V * 4 && H <= kThreads * kLnMaxIt) {
-      // wave-per-row kernel: 4 rows in flight per workgroup
-      constexpr int kWaves = kThreads / 64;
-      const long wg = (nrows + kWaves - 1) / kWaves;
-      grid = static_cast<int>(wg < kLnBwdMaxGrid ? (wg > 0 ? wg : 1)
-                                                 : kLnBwdMaxGrid);
-      const size_t lds = 2 * static_cast<size_t>(H) * sizeof(float);
-      const int nit = (H + 64 * V - 1) / (64 * V);
-#define BF_LAUNCH_LN_BWDW(NIT)                                                \
-  hipLaunchKernelGGL((ln_add_bwd_wave_k<T, NIT, V>), dim3(grid),              \
-                     dim3(kThreads), lds, stream, static_cast<T*>(dx),        \
-                     static_cast<const T*>(x), static_cast<const T*>(r),      \
-                     static_cast<const T*>(dy), gamma, mean, rstd, scratch,   \
-                     H, nrows)
-      if (nit <= 1) {
-        BF_LAUNCH_LN_BWDW(1);
-      } else if (nit <= 2) {
-        BF_LAUNCH_LN_BWDW(2);
-      } else if (nit <= 3) {
-        BF_LAUNCH_LN_BWDW(3);
-      } else {
-        BF_LAUNCH_LN_BWDW(4);
-      }
-#undef BF_LAUNCH_LN_BWDW
-    } else {
-      grid = ln_bwd_grid(nrows);
+    const int grid = ln_bwd_grid(nrows);
 #define BF_LAUNCH_LN_BWD(NIT, VV)                                             \
   hipLaunchKernelGGL((ln_add_bwd_k<T, NIT, VV>), dim3(grid), dim3(kThreads),  \
                      0, stream, static_cast<T*>(dx), static_cast<const T*>(x),\
                      static_cast<const T*>(r), static_cast<const T*>(dy),     \
                      gamma, mean, rstd, scratch, H, nrows)
-      BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
+    BF_LN_DISPATCH(BF_LAUNCH_LN_BWD);
 #undef BF_LAUNCH_LN_BWD
-    }
     BF_CHECK_HIP(hipGetLastError());
     hipLaunchKernelGGL(ln_bwd_finalize_k,
                        dim3(((H + 255) / 256) * kLnFinalizeSlots), dim3(256),
@@ -1551,30 +972,6 @@ hipError_t bf_add_relu_fwd(void* out, const void* a, const void* b,
 hipError_t bf_relu_bwd_mask(void* gin, const void* g, const void* out,
                             long numel, int dtype, hipStream_t stream) {
   return dispatch_dtype<ReluBwdMaskLauncher>(dtype, gin, g, out, numel, stream);
-}
-
-hipError_t bf_bn_relu_fwd(void* y, const void* x, const float* gamma,
-                          const float* beta, const float* run_mean_in,
-                          const float* run_var_in, float* run_mean,
-                          float* run_var, float* save_mean, float* save_invstd,
-                          float* ws, float momentum, float eps, int C, long HW,
-                          long numel, bool nhwc, int dtype,
-                          hipStream_t stream) {
-  return dispatch_dtype<BnFwdLauncher>(dtype, y, x, gamma, beta, run_mean_in,
-                                       run_var_in, run_mean, run_var, save_mean,
-                                       save_invstd, ws, momentum, eps, C, HW,
-                                       numel, nhwc, stream);
-}
-
-hipError_t bf_bn_relu_bwd(void* dx, const void* x, const void* dy,
-                          const float* gamma, const float* beta,
-                          const float* save_mean, const float* save_invstd,
-                          float* ws, float* dgamma, float* dbeta, float* coef,
-                          int C, long HW, long numel, bool nhwc, int dtype,
-                          hipStream_t stream) {
-  return dispatch_dtype<BnBwdLauncher>(dtype, dx, x, dy, gamma, beta, save_mean,
-                                       save_invstd, ws, dgamma, dbeta, coef, C,
-                                       HW, numel, nhwc, stream);
 }
 
 hipError_t bf_ln_add_fwd(void* y, const void* x, const void* r,

@@ -9,7 +9,7 @@ from typing import List, Type, Union
 import torch
 import torch.nn as nn
 
-from bluefog_amd.ops.fused_modules import FusedBNReLU2d, add_relu
+from bluefog_amd.ops.fused_modules import add_relu
 
 __all__ = ["ResNet", "resnet18", "resnet34", "resnet50", "resnet101", "resnet152"]
 
@@ -20,14 +20,15 @@ class BasicBlock(nn.Module):
     def __init__(self, in_planes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 3, stride, 1, bias=False)
-        self.bn1 = FusedBNReLU2d(planes)
+        self.bn1 = nn.BatchNorm2d(planes)
         self.conv2 = nn.Conv2d(planes, planes, 3, 1, 1, bias=False)
         self.bn2 = nn.BatchNorm2d(planes)
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
-        out = self.bn1(self.conv1(x))  # fused BN+ReLU
+        # in-place ReLU on the BN output (no other consumer)
+        out = torch.relu_(self.bn1(self.conv1(x)))
         out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -40,17 +41,17 @@ class Bottleneck(nn.Module):
     def __init__(self, in_planes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 1, 1, 0, bias=False)
-        self.bn1 = FusedBNReLU2d(planes)
+        self.bn1 = nn.BatchNorm2d(planes)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride, 1, bias=False)
-        self.bn2 = FusedBNReLU2d(planes)
+        self.bn2 = nn.BatchNorm2d(planes)
         self.conv3 = nn.Conv2d(planes, planes * 4, 1, 1, 0, bias=False)
         self.bn3 = nn.BatchNorm2d(planes * 4)
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
-        out = self.bn1(self.conv1(x))  # fused BN+ReLU
-        out = self.bn2(self.conv2(out))
+        out = torch.relu_(self.bn1(self.conv1(x)))
+        out = torch.relu_(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
         if self.downsample is not None:
             identity = self.downsample(x)
@@ -68,7 +69,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, 2, 3, bias=False)
-        self.bn1 = FusedBNReLU2d(64)
+        self.bn1 = nn.BatchNorm2d(64)
         self.maxpool = nn.MaxPool2d(3, 2, 1)
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], 2)
@@ -105,7 +106,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.bn1(self.conv1(x)))  # fused BN+ReLU
+        x = self.maxpool(torch.relu_(self.bn1(self.conv1(x))))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

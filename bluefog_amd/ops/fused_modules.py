@@ -16,15 +16,6 @@ import torch
 from bluefog_amd.ops import hip_ext
 
 
-def fused_bn_enabled() -> bool:
-    """Fused BN+ReLU is opt-in (BLUEFOG_FUSED_BN=1): numerically verified
-    (tests/test_gpu_kernels.py) but round-1 measurements show MIOpen's tuned
-    BN chain is still faster end-to-end on ResNet50 — see
-    profiles/ notes; kept for further kernel tuning."""
-    return os.environ.get("BLUEFOG_FUSED_BN", "0") == "1"
-
-
-
 def _dense(t: torch.Tensor) -> bool:
     return t.is_contiguous() or t.is_contiguous(memory_format=torch.channels_last)
 
@@ -71,68 +62,6 @@ def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     ):
         return _AddReLU.apply(a, b)
     return torch.relu(a + b)
-
-
-# ---------------------------------------------------------------------------
-# fused training BatchNorm2d + ReLU
-# ---------------------------------------------------------------------------
-
-
-def _bn_fusable(x: torch.Tensor, weight) -> bool:
-    return (
-        x.is_cuda
-        and x.dim() == 4
-        and x.dtype in (torch.float32, torch.float16, torch.bfloat16)
-        and weight is not None
-        and weight.dtype == torch.float32
-        and (x.is_contiguous() or x.is_contiguous(memory_format=torch.channels_last))
-        and x.size(1) <= 4096  # kernel's per-workgroup LDS budget
-        and hip_ext.has_extension()
-    )
-
-
-class _BNReLU(torch.autograd.Function):
-    @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps, ws):
-        from bluefog_amd import _C
-
-        C = x.size(1)
-        y = torch.empty_like(x)
-        f32 = dict(device=x.device, dtype=torch.float32)
-        save_mean = torch.empty(C, **f32)
-        save_invstd = torch.empty(C, **f32)
-        # ws is the module's persistent [2C] fp32 workspace; the finalize
-        # kernels re-zero it after reading, so it is always clean here
-        _C.bn_relu_fwd(
-            y, x, weight, bias, running_mean, running_var,
-            save_mean, save_invstd, ws, float(momentum), float(eps),
-        )
-        ctx.save_for_backward(x, weight, bias, save_mean, save_invstd)
-        ctx.ws = ws
-        return y
-
-    @staticmethod
-    def backward(ctx, grad_out):
-        from bluefog_amd import _C
-
-        x, weight, bias, save_mean, save_invstd = ctx.saved_tensors
-        mf = (
-            torch.channels_last
-            if x.is_contiguous(memory_format=torch.channels_last)
-            else torch.contiguous_format
-        )
-        grad_out = grad_out.contiguous(memory_format=mf)
-        C = x.size(1)
-        f32 = dict(device=x.device, dtype=torch.float32)
-        dx = torch.empty_like(x)
-        dgamma = torch.empty(C, **f32)
-        dbeta = torch.empty(C, **f32)
-        coef = torch.empty(3 * C, **f32)
-        _C.bn_relu_bwd(
-            dx, x, grad_out, weight, bias, save_mean, save_invstd,
-            ctx.ws, dgamma, dbeta, coef,
-        )
-        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 class _AddLayerNorm(torch.autograd.Function):
@@ -203,42 +132,3 @@ class FusedAddLayerNorm(torch.nn.LayerNorm):
         if len(self.normalized_shape) == 1 and _ln_fusable(x, residual, self.weight):
             return _AddLayerNorm.apply(x, residual, self.weight, self.bias, self.eps)
         return super().forward(x + residual)
-
-
-class FusedBNReLU2d(torch.nn.BatchNorm2d):
-    """BatchNorm2d immediately followed by ReLU, fused into gfx950 kernels
-    in training on GPU: stats (one read of x) -> normalize+ReLU (one
-    read/write), and a backward whose ReLU mask is recomputed from x —
-    replacing MIOpen's BN kernel chain plus torch's separate clamp /
-    threshold_backward passes. State-dict compatible with BatchNorm2d;
-    falls back to eager BN+ReLU in eval mode, on CPU, or for unsupported
-    layouts/dtypes.
-    """
-
-    def forward(self, x):
-        if (
-            fused_bn_enabled()
-            and self.training
-            and self.affine
-            and self.track_running_stats
-            and _bn_fusable(x, self.weight)
-        ):
-            if self.num_batches_tracked is not None:
-                self.num_batches_tracked.add_(1)
-            if self.momentum is None:
-                momentum = 1.0 / float(self.num_batches_tracked)
-            else:
-                momentum = self.momentum
-            ws = getattr(self, "_bf_ws", None)
-            if ws is None or ws.device != x.device:
-                ws = torch.zeros(
-                    2 * self.num_features, device=x.device, dtype=torch.float32
-                )
-                self._bf_ws = ws
-            return _BNReLU.apply(
-                x, self.weight, self.bias, self.running_mean, self.running_var,
-                momentum, self.eps, ws,
-            )
-        # eager fallback: in-place ReLU on the BN output (the pre-fusion
-        # fast path — BN output has no other consumer)
-        return torch.relu_(super().forward(x))

@@ -241,80 +241,6 @@ def test_add_relu_gpu_matches_eager(dev, dtype, channels_last):
     assert torch.equal(b.grad, b2.grad)
 
 
-@pytest.mark.parametrize("channels_last", [False, True])
-@pytest.mark.parametrize("dtype,atol", [(torch.float32, 3e-4), (torch.bfloat16, 6e-2)])
-def test_fused_bn_relu_matches_torch(dev, channels_last, dtype, atol, monkeypatch):
-    """FusedBNReLU2d (training) vs eager BatchNorm2d+ReLU: outputs, input
-    grads, parameter grads and running stats must agree."""
-    from bluefog_amd.ops import fused_modules
-    from bluefog_amd.ops.fused_modules import FusedBNReLU2d
-
-    monkeypatch.setenv("BLUEFOG_FUSED_BN", "1")
-    assert fused_modules.fused_bn_enabled()
-
-    torch.manual_seed(5)
-    N, C, H, W = 8, 32, 13, 9
-    mf = torch.channels_last if channels_last else torch.contiguous_format
-    fused = FusedBNReLU2d(C).to(dev)
-    ref = torch.nn.BatchNorm2d(C).to(dev)
-    ref.load_state_dict({k: v.clone() for k, v in fused.state_dict().items()})
-    # non-trivial affine params
-    with torch.no_grad():
-        for m in (fused, ref):
-            m.weight.uniform_(0.5, 1.5)
-            m.bias.uniform_(-0.5, 0.5)
-        ref.weight.copy_(fused.weight)
-        ref.bias.copy_(fused.bias)
-
-    x0 = (torch.randn(N, C, H, W, device=dev) * 2 + 0.7).to(dtype)
-    x1 = x0.clone().to(memory_format=mf).requires_grad_()
-    x2 = x0.clone().to(memory_format=mf).requires_grad_()
-    fused.train(), ref.train()
-    y1 = fused(x1)
-    y2 = torch.relu(ref(x2))
-    torch.cuda.synchronize()
-    assert torch.allclose(y1.float(), y2.float(), atol=atol), (
-        (y1.float() - y2.float()).abs().max().item()
-    )
-    g = torch.randn_like(y1)
-    y1.backward(g)
-    y2.backward(g.clone())
-    torch.cuda.synchronize()
-    assert torch.allclose(x1.grad.float(), x2.grad.float(), atol=atol), (
-        (x1.grad.float() - x2.grad.float()).abs().max().item()
-    )
-    assert torch.allclose(fused.weight.grad, ref.weight.grad, atol=5e-3 if dtype == torch.bfloat16 else 1e-2), (
-        (fused.weight.grad - ref.weight.grad).abs().max().item()
-    )
-    assert torch.allclose(fused.bias.grad, ref.bias.grad, atol=5e-3 if dtype == torch.bfloat16 else 1e-2), (
-        (fused.bias.grad - ref.bias.grad).abs().max().item()
-    )
-    assert torch.allclose(fused.running_mean, ref.running_mean, atol=1e-3), (
-        (fused.running_mean - ref.running_mean).abs().max().item()
-    )
-    assert torch.allclose(fused.running_var, ref.running_var, atol=1e-2), (
-        (fused.running_var - ref.running_var).abs().max().item()
-    )
-    assert int(fused.num_batches_tracked) == int(ref.num_batches_tracked)
-
-
-def test_fused_bn_relu_eval_matches_torch(dev):
-    from bluefog_amd.ops.fused_modules import FusedBNReLU2d
-
-    torch.manual_seed(6)
-    fused = FusedBNReLU2d(16).to(dev)
-    ref = torch.nn.BatchNorm2d(16).to(dev)
-    # push some stats through training first
-    fused.train()
-    for _ in range(3):
-        fused(torch.randn(4, 16, 8, 8, device=dev))
-    ref.load_state_dict(fused.state_dict())
-    fused.eval(), ref.eval()
-    x = torch.randn(4, 16, 8, 8, device=dev)
-    torch.cuda.synchronize()
-    assert torch.allclose(fused(x), torch.relu(ref(x)), atol=1e-5)
-
-
 @pytest.mark.parametrize(
     "dtype,atol,gatol",
     [
