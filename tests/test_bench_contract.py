@@ -116,6 +116,32 @@ def test_bench_config_presets():
         sys.argv = old
 
 
+def test_bench_torchrun_default_optimizer():
+    """The driver's SCALE tier launches bench.py under torch.distributed.run
+    with the DEFAULT optimizer; rehearse that exact invocation at ws=2."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(_ROOT, "bench.py"), "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--model", "resnet18", "--batch-size", "2",
+         "--device", "cpu"],
+        cwd=_ROOT, env=dict(os.environ), capture_output=True, text=True,
+        timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["config"]["global_batch"] == 4, rec
+    assert "dynamic one-peer exp2" in rec["config"]["parallelism"], rec
+
+
 def test_bench_hierarchical_faked_machines():
     """BASELINE config 5's code path end to end: 4 ranks, 2 faked machines
     (BLUEFOG_NODES_PER_MACHINE), hierarchical_neighbor_allreduce."""
