@@ -312,6 +312,55 @@ def test_push_sum_consistency():
     run_dist(w_push_sum_consistency, 4, timeout=300)
 
 
+def w_win_multiwindow_stress():
+    """Multi-window, multi-writer: 4 ranks, 3 windows each; every rank
+    accumulates into BOTH its ring neighbors' windows while every rank
+    concurrently collects its own — per (window, origin) totals must be
+    exact at the end (no lost updates, no torn reads across windows
+    sharing the single worker thread and the batched control plane)."""
+    import time
+
+    import bluefog_amd as bf
+
+    bf.init()
+    rank, size = bf.rank(), bf.size()
+    bf.set_topology(bf.RingGraph(size))
+    n_puts = 15
+    names = [f"mw{w}" for w in range(3)]
+    for name in names:
+        bf.win_create(torch.zeros(64), name, zero_init=True)
+    bf.barrier()
+    dsts = {r: 1.0 for r in bf.out_neighbor_ranks()}
+    ones = torch.ones(64)
+    for i in range(n_puts):
+        for name in names:
+            bf.win_accumulate(ones, name, dst_weights=dsts, require_mutex=True)
+        if i % 5 == 4:
+            for name in names:
+                # interleaved collects fold delivered values into the
+                # window tensor in place — racing the writers on purpose
+                bf.win_update_then_collect(name)
+    bf.barrier()  # all accumulates delivered (win_accumulate blocks)
+    deadline = time.time() + 120
+    expected = float(n_puts * len(bf.in_neighbor_ranks()))
+    for name in names:
+        out = bf.win_update_then_collect(name)
+        while time.time() < deadline and float(out[0]) < expected:
+            out = bf.win_update_then_collect(name)
+            time.sleep(0.001)
+        # a lost update undershoots forever; a torn/duplicated read
+        # overshoots — only the exact total passes
+        assert torch.allclose(out, torch.full((64,), expected)), (
+            name, float(out.min()), float(out.max()), expected,
+        )
+    bf.barrier()
+    bf.win_free()
+
+
+def test_win_multiwindow_stress():
+    run_dist(w_win_multiwindow_stress, 4, timeout=300)
+
+
 def w_win_mutex_stress():
     """Atomicity stress: rank 0 fires many win_accumulates while rank 1
     concurrently runs win_update_then_collect (which resets the buffer).
