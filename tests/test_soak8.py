@@ -133,3 +133,13 @@ def test_soak8_awc_dynamic_exp2():
 @pytest.mark.timeout(600)
 def test_soak8_win_put_rotation():
     run_dist(w_soak_win_put_rotation, 8, timeout=540)
+
+
+@pytest.mark.timeout(600)
+def test_soak8_awc_with_consistency_checker():
+    """The debug coordinator must stay silent on the legitimate dynamic
+    exp2 schedule (its per-rank src/dst sets differ — they are detail,
+    not digest) while adding bounded overhead."""
+    run_dist(w_soak_awc_dynamic_exp2, 8,
+             env={"BLUEFOG_CHECK_CONSISTENCY": "32",
+                  "BLUEFOG_SOAK_STEPS": "60"}, timeout=540)
