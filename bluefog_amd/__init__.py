@@ -163,6 +163,19 @@ def rccl_built() -> bool:
 # ---------------------------------------------------------------------------
 
 
+def check_extension(ext_name: str = "bluefog_amd._C", pkg_path=None, *args) -> None:
+    """Raise ImportError if the native gfx950 extension is not built
+    (reference analog: bluefog/common/util.py:47-52, called at package
+    import to fail early). Extra args are accepted for signature parity."""
+    from bluefog_amd.ops import hip_ext
+
+    if not hip_ext.has_extension():
+        raise ImportError(
+            f"Extension {ext_name} has not been built. Run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`."
+        )
+
+
 def timeline_start_activity(tensor_name: str, activity_name: str) -> bool:
     from bluefog_amd.utils.timeline import timeline
 
