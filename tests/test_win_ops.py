@@ -403,3 +403,26 @@ def w_win_mutex_stress():
 
 def test_win_mutex_stress():
     run_dist(w_win_mutex_stress, 2, timeout=180.0)
+
+
+def test_win_latency_probe_script():
+    """The dev probe (scripts/win_latency_bench.py) self-spawns 2 ranks
+    and prints a JSON report (reference analog: the standalone window
+    experiments under scripts/)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    for k in ("RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT", "LOCAL_RANK"):
+        env.pop(k, None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "scripts", "win_latency_bench.py"),
+         "--sizes", "512", "--iters", "3"],
+        capture_output=True, text=True, timeout=300, env=env, cwd=root,
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    rows = json.loads(out.stdout[out.stdout.index("["):])
+    assert rows and rows[0]["put_store_rpcs"] <= 2.0, rows
