@@ -424,5 +424,7 @@ def test_win_latency_probe_script():
         capture_output=True, text=True, timeout=300, env=env, cwd=root,
     )
     assert out.returncode == 0, out.stderr[-1500:]
-    rows = json.loads(out.stdout[out.stdout.index("["):])
+    # skip the "[Gloo] ..." banner lines; the report starts at a bare "["
+    start = out.stdout.index("[\n")
+    rows = json.loads(out.stdout[start:])
     assert rows and rows[0]["put_store_rpcs"] <= 2.0, rows
