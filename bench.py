@@ -37,7 +37,7 @@ _BASELINE_CONFIGS = {
     2: dict(model="resnet50", batch_size=64, dist_optimizer="neighbor_allreduce"),
     3: dict(model="resnet50", batch_size=32, dist_optimizer="neighbor_allreduce"),
     4: dict(model="resnet50", batch_size=64, dist_optimizer="win_put"),
-    5: dict(model="bert_base", seq_len=512, batch_size=16,
+    5: dict(model="bert_base", seq_len=512, batch_size=64,
             dist_optimizer="hierarchical_neighbor_allreduce"),
 }
 
