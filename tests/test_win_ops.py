@@ -160,6 +160,33 @@ def w_win_version():
     bf.win_free("wv")
 
 
+def w_win_version_get_path():
+    """win_get also counts as an update of the local buffer (split "get"
+    counter, single-writer = the owner): version must rise on get and
+    clear on win_update, mixed freely with incoming puts."""
+    bf = _init_ring()
+    rank = bf.rank()
+    t = torch.ones(2) * (rank + 1)
+    bf.win_create(t, "wvg")
+    bf.barrier()
+    src = bf.in_neighbor_ranks()[0]
+    bf.win_get("wvg", src_weights={src: 1.0})
+    ver = bf.get_win_version("wvg")
+    assert ver[src] == 1, ver
+    bf.barrier()
+    bf.win_put(t, "wvg")  # each out-neighbor's buffer for us bumps too
+    bf.barrier()
+    ver = bf.get_win_version("wvg")
+    # src slot saw our get (1) plus src's put (1); other in-neighbors saw
+    # only their put
+    assert ver[src] == 2, ver
+    assert all(v == 1 for r, v in ver.items() if r != src), ver
+    bf.win_update("wvg")
+    assert all(v == 0 for v in bf.get_win_version("wvg").values())
+    bf.barrier()
+    bf.win_free("wvg")
+
+
 def w_win_mutex():
     bf = _init_ring()
     rank, size = bf.rank(), bf.size()
@@ -294,6 +321,10 @@ def test_win_get():
 
 def test_win_version():
     run_dist(w_win_version, 2)
+
+
+def test_win_version_get_path():
+    run_dist(w_win_version_get_path, 4)
 
 
 def test_win_mutex():
