@@ -65,6 +65,47 @@ def _end_backward_span():
     timeline().end_activity("model.compute")
 
 
+
+def _combine_step_torch(kind, b, group, gathered, weights, self_weight):
+    """Torch-op replica of the fused combine+step kernels, used when the
+    bucket lives on CPU (BLUEFOG_FUSED_STEP=force): lets multi-rank CPU
+    soaks drive the EXACT fused control flow the 8-GPU run executes. Math
+    mirrors csrc weighted_combine_sgd/adam (verified against torch.optim
+    on GPU by tests/test_gpu_fused.py)."""
+    flat, grad = b["flat"], b["grad_flat"]
+    numel = flat.numel()
+    acc = flat.mul(float(self_weight))
+    for k, w in enumerate(weights):
+        acc.add_(gathered.narrow(0, k * numel, numel), alpha=float(w))
+    if kind == "sgd":
+        mu = group["momentum"]
+        wd = group["weight_decay"]
+        lr = group["lr"]
+        nesterov = group.get("nesterov", False)
+        d = grad.add(acc, alpha=wd) if wd != 0 else grad.clone()
+        if mu != 0:
+            if b["momentum_flat"] is None:
+                b["momentum_flat"] = torch.zeros_like(flat)
+            buf = b["momentum_flat"]
+            buf.mul_(mu).add_(d)
+            d = d.add(buf, alpha=mu) if nesterov else buf
+        acc.add_(d, alpha=-lr)
+    else:
+        b["adam_step"] += 1
+        t = b["adam_step"]
+        beta1, beta2 = group["betas"]
+        eps, lr, wd = group["eps"], group["lr"], group["weight_decay"]
+        g = grad.add(acc, alpha=wd) if wd != 0 else grad
+        m, v = b["exp_avg"], b["exp_avg_sq"]
+        m.mul_(beta1).add_(g, alpha=1 - beta1)
+        v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        bias1 = 1 - beta1 ** t
+        bias2 = 1 - beta2 ** t
+        denom = (v / bias2).sqrt_().add_(eps)
+        acc.addcdiv_(m / bias1, denom, value=-lr)
+    flat.copy_(acc)
+
+
 class CommunicationType(Enum):
     neighbor_allreduce = "neighbor.allreduce"
     hierarchical_neighbor_allreduce = "hierarchical.neighbor.allreduce"
@@ -426,16 +467,19 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
     # per-bucket buffers so the kernel sees one contiguous grad stream.
     # ------------------------------------------------------------------
     def _init_fused_mode(self):
-        if os.environ.get("BLUEFOG_FUSED_STEP", "1") in ("0", "false"):
+        mode = os.environ.get("BLUEFOG_FUSED_STEP", "1")
+        if mode in ("0", "false"):
             return None
         if not self._buckets.buckets:
             return None
         dev = self._buckets.buckets[0]["flat"].device
-        if dev.type != "cuda":
+        if dev.type != "cuda" and mode != "force":
+            # "force" drives the fused control flow with a torch-op step
+            # replica on CPU (multi-rank soak coverage of the GPU path)
             return None
         from bluefog_amd.ops import hip_ext
 
-        if not hip_ext.has_extension():
+        if dev.type == "cuda" and not hip_ext.has_extension():
             return None
         if len(self.param_groups) != 1:
             return None
@@ -482,10 +526,13 @@ class _DistributedReduceOptimizer(torch.optim.Optimizer):
         from bluefog_amd import _C
 
         for w in works:
-            w.wait()  # stream-level wait on RCCL work
+            w.wait()  # stream-level wait on RCCL work (CPU: host wait)
         group = self.param_groups[0]
         flat, grad = b["flat"], b["grad_flat"]
         gathered = gathered if gathered is not None else flat
+        if not flat.is_cuda:
+            return _combine_step_torch(self._fused, b, group, gathered,
+                                       weights, self_weight)
         if self._fused == "sgd":
             momentum = group["momentum"]
             if momentum != 0 and b["momentum_flat"] is None:
@@ -736,16 +783,19 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
         self._register_hooks()
 
     def _init_fused_mode(self):
-        if os.environ.get("BLUEFOG_FUSED_STEP", "1") in ("0", "false"):
+        mode = os.environ.get("BLUEFOG_FUSED_STEP", "1")
+        if mode in ("0", "false"):
             return None
         if not self._buckets.buckets:
             return None
         dev = self._buckets.buckets[0]["flat"].device
-        if dev.type != "cuda":
+        if dev.type != "cuda" and mode != "force":
+            # "force" drives the fused control flow with a torch-op step
+            # replica on CPU (multi-rank soak coverage of the GPU path)
             return None
         from bluefog_amd.ops import hip_ext
 
-        if not hip_ext.has_extension():
+        if dev.type == "cuda" and not hip_ext.has_extension():
             return None
         if len(self.param_groups) != 1:
             return None
@@ -792,6 +842,9 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
 
         group = self.param_groups[0]
         flat, grad = b["flat"], b["grad_flat"]
+        if not flat.is_cuda:
+            _combine_step_torch(self._fused, b, group, flat, [], 1.0)
+            return self._fused_bucket_exchange(b)
         if self._fused == "sgd":
             momentum = group["momentum"]
             if momentum != 0 and b["momentum_flat"] is None:
@@ -813,6 +866,9 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
                 float(group["eps"]), float(group["weight_decay"]),
                 int(b["adam_step"]),
             )
+        self._fused_bucket_exchange(b)
+
+    def _fused_bucket_exchange(self, b):
         if (
             _bf().size() > 1
             and self._communication_type == CommunicationType.neighbor_allreduce
@@ -822,7 +878,7 @@ class _DistributedAdaptThenCombineOptimizer(torch.optim.Optimizer):
             self._handles[b["name"]] = (
                 "fused",
                 *neighbor.post_neighbor_exchange_raw(
-                    flat,
+                    b["flat"],
                     self.self_weight,
                     self.src_weights,
                     self.dst_weights,

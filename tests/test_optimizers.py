@@ -460,3 +460,67 @@ def test_duplicated_module():
 
 def test_unused_head_flush():
     run_dist(w_unused_head_flush, 2, timeout=300)
+
+
+def w_awc_fused_forced_matches_plain():
+    """BLUEFOG_FUSED_STEP=force (CPU torch replica of the fused kernels)
+    must train identically to plain torch SGD at world size 1 — the same
+    equivalence the GPU kernels prove in tests/test_gpu_fused.py."""
+    import copy
+
+    import bluefog_amd as bf
+
+    bf.init()
+    torch.manual_seed(31)
+    m_ref = nn.Sequential(nn.Linear(24, 48), nn.ReLU(), nn.Linear(48, 8))
+    m_fused = copy.deepcopy(m_ref)
+    opt_ref = torch.optim.SGD(m_ref.parameters(), lr=0.05, momentum=0.9,
+                              weight_decay=1e-4)
+    opt_fused = bf.DistributedAdaptWithCombineOptimizer(
+        torch.optim.SGD(m_fused.parameters(), lr=0.05, momentum=0.9,
+                        weight_decay=1e-4),
+        model=m_fused,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+    )
+    assert opt_fused._fused == "sgd", "forced fused mode must engage"
+    lf = nn.MSELoss()
+    torch.manual_seed(77)
+    for _ in range(6):
+        x = torch.randn(16, 24)
+        y = torch.randn(16, 8)
+        opt_ref.zero_grad()
+        lf(m_ref(x), y).backward()
+        opt_ref.step()
+        opt_fused.zero_grad()
+        lf(m_fused(x), y).backward()
+        opt_fused.step()
+    for a, b in zip(m_ref.parameters(), m_fused.parameters()):
+        assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max().item()
+
+
+def test_awc_fused_forced_matches_plain():
+    run_dist(w_awc_fused_forced_matches_plain, 1,
+             env={"BLUEFOG_FUSED_STEP": "force"}, timeout=300)
+
+
+def w_atc_fused_forced():
+    """ATC with the forced CPU fused replica converges at ws=2."""
+    import bluefog_amd as bf
+
+    bf.init()
+    bf.set_topology(bf.ExponentialTwoGraph(bf.size()))
+    problem = _Problem(bf.rank())
+    model = _make_model()
+    opt = bf.DistributedAdaptThenCombineOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.05),
+        model=model,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+    )
+    assert opt._fused == "sgd", "forced fused mode must engage"
+    _train(bf, opt, model, problem, 200)
+    _assert_converged(bf, model, problem, 5e-3)
+
+
+def test_atc_fused_forced():
+    run_dist(w_atc_fused_forced, 2,
+             env={"BLUEFOG_FUSED_STEP": "force"}, timeout=300)

@@ -74,6 +74,10 @@ def w_soak_awc_dynamic_exp2():
         model=model,
         communication_type=bf.CommunicationType.neighbor_allreduce,
     )
+    import os as _os
+
+    if _os.environ.get("BLUEFOG_FUSED_STEP") == "force":
+        assert opt._fused == "sgd", "forced fused mode must engage on CPU"
     gen = tu.GetDynamicOnePeerSendRecvRanks(topo, bf.rank())
     bf.broadcast_parameters(model.state_dict(), root_rank=0)
     x = torch.randn(8, 32)
@@ -143,3 +147,14 @@ def test_soak8_awc_with_consistency_checker():
     run_dist(w_soak_awc_dynamic_exp2, 8,
              env={"BLUEFOG_CHECK_CONSISTENCY": "32",
                   "BLUEFOG_SOAK_STEPS": "60"}, timeout=540)
+
+
+@pytest.mark.timeout(600)
+def test_soak8_awc_fused_forced():
+    """The EXACT 8-GPU flagship combination — fused bucket exchange +
+    fused combine/step + dynamic one-peer exp2 — soaked at 8 ranks on
+    CPU via the torch-op replica of the fused kernels
+    (BLUEFOG_FUSED_STEP=force)."""
+    run_dist(w_soak_awc_dynamic_exp2, 8,
+             env={"BLUEFOG_FUSED_STEP": "force",
+                  "BLUEFOG_SOAK_STEPS": "120"}, timeout=540)
