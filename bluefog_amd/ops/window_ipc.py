@@ -66,6 +66,10 @@ class IpcWindowPeers:
         self._peer_slot: Dict[int, torch.Tensor] = {}
         for dst in win.out_ranks:
             block = ipc_open(store.get(f"win/{name}/ipc/{dst}"))
+            # slot layout contract: the destination built its block in
+            # ITS ctx.in_neighbor_ranks() order, which set_topology defines
+            # as sorted(predecessors) — recompute the same sorted list here
+            # so our slot index matches the peer's layout exactly
             dst_in_ranks = sorted(r for r in topo.predecessors(dst) if r != dst)
             idx = dst_in_ranks.index(me)
             self._peer_slot[dst] = block[idx].view(win.self_tensor.shape)
