@@ -524,3 +524,57 @@ def w_atc_fused_forced():
 def test_atc_fused_forced():
     run_dist(w_atc_fused_forced, 2,
              env={"BLUEFOG_FUSED_STEP": "force"}, timeout=300)
+
+
+def w_awc_fused_vs_nonfused_dst_weighted():
+    """Fused and non-fused AWC must produce identical training given the
+    same dynamic schedule WITH non-unit dst weights (covers the weighted
+    send-copy branch of the raw bucket exchange). Both optimizers live in
+    one world; program order is identical on every rank."""
+    import os as _os
+
+    import bluefog_amd as bf
+    import bluefog_amd.parallel.topology as tu
+
+    bf.init()
+    topo = bf.ExponentialTwoGraph(bf.size())
+    bf.set_topology(topo)
+    torch.manual_seed(4321)
+    m_fused = nn.Linear(16, 4, bias=False)
+    torch.manual_seed(4321)
+    m_plain = nn.Linear(16, 4, bias=False)
+    _os.environ["BLUEFOG_FUSED_STEP"] = "force"
+    opt_fused = bf.DistributedAdaptWithCombineOptimizer(
+        torch.optim.SGD(m_fused.parameters(), lr=0.03),
+        model=m_fused,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+    )
+    _os.environ["BLUEFOG_FUSED_STEP"] = "0"
+    opt_plain = bf.DistributedAdaptWithCombineOptimizer(
+        torch.optim.SGD(m_plain.parameters(), lr=0.03),
+        model=m_plain,
+        communication_type=bf.CommunicationType.neighbor_allreduce,
+    )
+    assert opt_fused._fused == "sgd" and opt_plain._fused is None
+    gen = tu.GetDynamicOnePeerSendRecvRanks(topo, bf.rank())
+    g = torch.Generator().manual_seed(5 + bf.rank())
+    for _ in range(25):
+        send, recv = next(gen)
+        w = 1.0 / (len(recv) + 1)
+        x = torch.randn(8, 16, generator=g)
+        for opt, m in ((opt_fused, m_fused), (opt_plain, m_plain)):
+            opt.self_weight = w
+            opt.src_weights = {r: w for r in recv}
+            # non-unit dst weights engage the weighted send-copy path
+            opt.dst_weights = {r: 0.5 for r in send}
+            opt.enable_topo_check = False
+            opt.zero_grad()
+            (m(x) ** 2).mean().backward()
+            opt.step()
+    a = m_fused.weight.detach()
+    b = m_plain.weight.detach()
+    assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max().item()
+
+
+def test_awc_fused_vs_nonfused_dst_weighted():
+    run_dist(w_awc_fused_vs_nonfused_dst_weighted, 4, timeout=300)
