@@ -371,6 +371,11 @@ class BlueFogContext:
     # ------------------------------------------------------------------
     # process groups & streams
     # ------------------------------------------------------------------
+    def machine_rank_list(self, machine: int) -> List[int]:
+        """Global ranks of a machine, in local-rank order (leader first)."""
+        self._require_init()
+        return list(self._machine_rank_lists[machine])
+
     def ensure_local_groups(self) -> None:
         """Build the per-machine ("local") and same-local-rank ("cross")
         process groups used by hierarchical ops. Collective: every rank must

@@ -413,9 +413,13 @@ def hierarchical_neighbor_allreduce_nonblocking(
     c.ensure_local_groups()
     local_size = c.local_size()
     local_rank = c.local_rank()
-    # machine m's leader is its first rank
+
+    # machine m's leader = its first rank in local-rank order. With faked
+    # machines (BLUEFOG_NODES_PER_MACHINE) that is m*local_size, but a
+    # hostname census may place a machine's ranks non-contiguously — use
+    # the actual census list, which is what the cross groups are built on
     def leader_of(machine: int) -> int:
-        return machine * local_size
+        return c.machine_rank_list(machine)[0]
 
     tensor = tensor.detach()
     buf = tensor.contiguous().clone()
