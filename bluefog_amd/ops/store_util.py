@@ -96,6 +96,10 @@ class ControlStore:
     # -- distributed ticket mutex -----------------------------------------
     # Fair FIFO lock: acquire takes a ticket (atomic add) and spins until the
     # serving counter reaches it. Replaces the reference's MPI RMA spin lock.
+    # A waiter that times out marks its ticket ABANDONED; when serving
+    # reaches an abandoned ticket, exactly one live waiter advances past it
+    # (guarded by a claim counter) — a timeout therefore never wedges the
+    # queue for everyone behind it.
     def mutex_acquire(self, name: str, timeout_s: float = 60.0) -> None:
         ticket = self.add(f"mutex/{name}/next", 1) - 1
         deadline = time.monotonic() + timeout_s
@@ -103,10 +107,20 @@ class ControlStore:
             serving = self.counter(f"mutex/{name}/serving")
             if serving == ticket:
                 return
+            if serving < ticket and self.check([f"mutex/{name}/abandon/{serving}"]):
+                # skip an abandoned turn; the claim counter makes the
+                # advance exactly-once even with many waiters
+                if self.add(f"mutex/{name}/skipped/{serving}", 1) == 1:
+                    self.add(f"mutex/{name}/serving", 1)
+                continue
             if time.monotonic() > deadline:
+                self.set(f"mutex/{name}/abandon/{ticket}", b"1")
+                # our turn may have arrived in the same instant: do NOT
+                # release it ourselves (a waiter could concurrently skip it
+                # and double-advance); the skip path retires it safely
                 raise TimeoutError(
                     f"bluefog_amd: timed out acquiring distributed mutex {name!r} "
-                    f"(ticket {ticket}, serving {serving})"
+                    f"(ticket {ticket}, serving {serving}); ticket abandoned"
                 )
             time.sleep(self.MUTEX_POLL_S)
 
